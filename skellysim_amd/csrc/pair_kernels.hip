@@ -1,0 +1,361 @@
+/* pair_kernels.hip — hand-written CDNA4 (gfx950/MI355X) fp64 pair kernels for
+ * SkellySim's hydrodynamic hot path.
+ *
+ * Built from scratch for MI355X; the math restates the reference
+ * (flatironinstitute/SkellySim) formula by formula:
+ *   Stokeslet            src/core/kernels.cu:62-76
+ *   Stresslet (DL)       src/core/kernels.cu:29-54
+ *   Regularized Oseen    src/core/kernels.cpp:85-131
+ *   Rotlet               src/core/kernels.cpp:206-242
+ *
+ * Design (nothing shared with the reference's CUDA driver, kernels.cu:80-123,
+ * which used 32-thread blocks and accumulated into global memory every tile):
+ *   - 256-thread blocks (4 wave64s), TPT targets per thread held in VGPRs;
+ *     velocity accumulates in registers and is written once.
+ *   - Sources staged through LDS in TILE=512-point tiles (24 KB stokeslet,
+ *     48 KB stresslet). Inner-loop LDS reads are wave-uniform -> broadcast,
+ *     conflict-free by construction.
+ *   - fp64 rsqrt: raw v_rsq_f64 + one 5-op Householder step (the same
+ *     refinement ROCm's libm applies after v_rsq_f64), skipping libm's
+ *     fp-class edge handling because r^2 is a finite sum of squares and the
+ *     r^2==0 case is masked explicitly (reference masks it too,
+ *     kernels.cu:39,70).
+ *   - Accumulation order per target is source order (tile-major), fixed ->
+ *     bit-reproducible for a given shard layout.
+ *
+ * Roofline: compute-bound on the fp64 VALU. The kernels are
+ * deliberately not GEMM-shaped: gfx950's fp64 matrix rate equals its vector
+ * rate, so MFMA buys nothing here.
+ */
+
+#include <hip/hip_runtime.h>
+
+#define BLOCK 256
+#define TILE 512
+
+/* v_rsq_f64 + one Householder(2nd order) refinement.
+ * v_rsq_f64 gives ~2^-26 relative error; the cubic-convergence step below
+ * (e = 1 - x*y^2; y += y*e*(0.5 + 0.375*e)) lands at <=2 ulp of fp64, which
+ * is the same polish ROCm libm's rsqrt(double) performs. Inputs here are
+ * finite and >= 0; x == 0 must be masked by the caller. */
+__device__ inline double rsq_refined(double x) {
+    double y;
+    asm("v_rsq_f64 %0, %1" : "=v"(y) : "v"(x));
+    const double e = __builtin_fma(-x * y, y, 1.0);
+    const double c = __builtin_fma(e, 0.375, 0.5);
+    return __builtin_fma(y * e, c, y);
+}
+
+/* ---- kernel functors ------------------------------------------------- */
+
+struct Stokeslet {
+    static constexpr int SRCDIM = 3;
+    struct Params {
+        double scale; /* 1/(8*pi) [ / eta when folded] */
+    };
+    /* u += (1/r)(f + rhat (f.rhat)); r = t - s; r==0 -> 0 (kernels.cu:62-76) */
+    __device__ static inline void pair(const double t[3], const double sp[3], const double f[3],
+                                       double acc[3], const Params &) {
+        const double dx = t[0] - sp[0];
+        const double dy = t[1] - sp[1];
+        const double dz = t[2] - sp[2];
+        double r2 = dx * dx;
+        r2 = __builtin_fma(dy, dy, r2);
+        r2 = __builtin_fma(dz, dz, r2);
+        double rinv = rsq_refined(r2);
+        rinv = (r2 == 0.0) ? 0.0 : rinv; /* kernels.cu:70 */
+        const double rinv2 = rinv * rinv;
+        double fdotr = f[0] * dx;
+        fdotr = __builtin_fma(f[1], dy, fdotr);
+        fdotr = __builtin_fma(f[2], dz, fdotr);
+        const double inner = fdotr * rinv2;
+        acc[0] = __builtin_fma(rinv, __builtin_fma(dx, inner, f[0]), acc[0]);
+        acc[1] = __builtin_fma(rinv, __builtin_fma(dy, inner, f[1]), acc[1]);
+        acc[2] = __builtin_fma(rinv, __builtin_fma(dz, inner, f[2]), acc[2]);
+    }
+    __device__ static inline double finish(double a, const Params &p) { return a * p.scale; }
+};
+
+struct Stresslet {
+    static constexpr int SRCDIM = 9;
+    struct Params {
+        double scale; /* 1/(8*pi) [ / eta when folded] */
+    };
+    /* u += -3 (d^T S d)/r^5 d; d = t - s; r==0 -> 0 (kernels.cu:29-54) */
+    __device__ static inline void pair(const double t[3], const double sp[3], const double f[9],
+                                       double acc[3], const Params &) {
+        const double dx = t[0] - sp[0];
+        const double dy = t[1] - sp[1];
+        const double dz = t[2] - sp[2];
+        const double dxx = dx * dx, dyy = dy * dy, dzz = dz * dz;
+        double dr2 = dxx + dyy;
+        dr2 += dzz;
+        const double rinv = rsq_refined(dr2);
+        const double rinv2 = rinv * rinv;
+        double rinv5 = rinv * rinv2 * rinv2;
+        rinv5 = (dr2 == 0.0) ? 0.0 : rinv5; /* kernels.cu:39 */
+        /* coeff = sxx dx^2 + syy dy^2 + szz dz^2 + (sxy+syx) dx dy
+         *       + (sxz+szx) dx dz + (syz+szy) dy dz   (kernels.cu:45-48) */
+        double coeff = f[0] * dxx;
+        coeff = __builtin_fma(f[4], dyy, coeff);
+        coeff = __builtin_fma(f[8], dzz, coeff);
+        coeff = __builtin_fma(f[1] + f[3], dx * dy, coeff);
+        coeff = __builtin_fma(f[2] + f[6], dx * dz, coeff);
+        coeff = __builtin_fma(f[5] + f[7], dy * dz, coeff);
+        coeff *= -3.0 * rinv5;
+        acc[0] = __builtin_fma(dx, coeff, acc[0]);
+        acc[1] = __builtin_fma(dy, coeff, acc[1]);
+        acc[2] = __builtin_fma(dz, coeff, acc[2]);
+    }
+    __device__ static inline double finish(double a, const Params &p) { return a * p.scale; }
+};
+
+struct OseenContract {
+    static constexpr int SRCDIM = 3;
+    struct Params {
+        double scale; /* factor = 1/(8*pi*eta), hoisted out of the pair loop */
+        double reg2;
+        double eps2;
+    };
+    /* kernels.cpp:96-127: dr==0 skipped; dr > eps -> fr=1/dr, gr=1/dr^3;
+     * else fr=1/sqrt(dr^2+reg^2), gr that cubed. The dr > eps comparison is
+     * evaluated as dr2 > eps^2 (monotone equivalent for exact values; can
+     * differ only when sqrt rounding straddles eps itself). `factor` is
+     * applied once at the end (linear in the sum). */
+    __device__ static inline void pair(const double t[3], const double sp[3], const double rho[3],
+                                       double acc[3], const Params &p) {
+        const double dx = sp[0] - t[0]; /* src - trg, kernels.cpp:99-101 */
+        const double dy = sp[1] - t[1];
+        const double dz = sp[2] - t[2];
+        double dr2 = dx * dx;
+        dr2 = __builtin_fma(dy, dy, dr2);
+        dr2 = __builtin_fma(dz, dz, dr2);
+        const double denom2 = (dr2 > p.eps2) ? dr2 : dr2 + p.reg2;
+        double y = rsq_refined(denom2);
+        y = (dr2 == 0.0) ? 0.0 : y; /* kernels.cpp:105-106 skip */
+        const double y3 = y * (y * y);
+        double ddotrho = dx * rho[0];
+        ddotrho = __builtin_fma(dy, rho[1], ddotrho);
+        ddotrho = __builtin_fma(dz, rho[2], ddotrho);
+        const double c = y3 * ddotrho;
+        acc[0] = __builtin_fma(y, rho[0], acc[0]);
+        acc[1] = __builtin_fma(y, rho[1], acc[1]);
+        acc[2] = __builtin_fma(y, rho[2], acc[2]);
+        acc[0] = __builtin_fma(dx, c, acc[0]);
+        acc[1] = __builtin_fma(dy, c, acc[1]);
+        acc[2] = __builtin_fma(dz, c, acc[2]);
+    }
+    __device__ static inline double finish(double a, const Params &p) { return a * p.scale; }
+};
+
+struct Rotlet {
+    static constexpr int SRCDIM = 3;
+    struct Params {
+        double scale; /* factor = 1/(8*pi*eta), applied at the end as the
+                         reference does (kernels.cpp:239) */
+        double reg2;
+        double eps2;
+    };
+    /* kernels.cpp:218-236: dr = dr2 < eps^2 ? sqrt(reg2+dr2) : sqrt(dr2)
+     * (no dr==0 skip); fr = 1/dr^3; u += fr * (d x' rho) with the reference's
+     * sign convention u_x += fr*(dz*rho_y - dy*rho_z), d = trg - src. */
+    __device__ static inline void pair(const double t[3], const double sp[3], const double rho[3],
+                                       double acc[3], const Params &p) {
+        const double dx = t[0] - sp[0];
+        const double dy = t[1] - sp[1];
+        const double dz = t[2] - sp[2];
+        double dr2 = dx * dx;
+        dr2 = __builtin_fma(dy, dy, dr2);
+        dr2 = __builtin_fma(dz, dz, dr2);
+        const double denom2 = (dr2 < p.eps2) ? dr2 + p.reg2 : dr2;
+        const double y = rsq_refined(denom2);
+        const double fr = y * (y * y);
+        const double cx = __builtin_fma(dz, rho[1], -(dy * rho[2]));
+        const double cy = __builtin_fma(dx, rho[2], -(dz * rho[0]));
+        const double cz = __builtin_fma(dy, rho[0], -(dx * rho[1]));
+        acc[0] = __builtin_fma(fr, cx, acc[0]);
+        acc[1] = __builtin_fma(fr, cy, acc[1]);
+        acc[2] = __builtin_fma(fr, cz, acc[2]);
+    }
+    __device__ static inline double finish(double a, const Params &p) { return a * p.scale; }
+};
+
+/* ---- driver ----------------------------------------------------------- */
+
+template <typename K, int TPT>
+__global__ __launch_bounds__(BLOCK) void pair_driver(const double *__restrict__ r_src,
+                                                     const double *__restrict__ f_src,
+                                                     const double *__restrict__ r_trg,
+                                                     double *__restrict__ u_trg,
+                                                     long long n_src, long long n_trg,
+                                                     typename K::Params params) {
+    __shared__ double lds[TILE * (3 + K::SRCDIM)];
+    double *lds_r = lds;
+    double *lds_f = lds + TILE * 3;
+    const int tid = threadIdx.x;
+    const long long base = (long long)blockIdx.x * (BLOCK * TPT);
+
+    double tp[TPT][3];
+    double acc[TPT][3];
+#pragma unroll
+    for (int k = 0; k < TPT; ++k) {
+        long long it = base + (long long)k * BLOCK + tid;
+        const long long itc = it < n_trg ? it : (n_trg > 0 ? n_trg - 1 : 0);
+        tp[k][0] = r_trg[3 * itc + 0];
+        tp[k][1] = r_trg[3 * itc + 1];
+        tp[k][2] = r_trg[3 * itc + 2];
+        acc[k][0] = acc[k][1] = acc[k][2] = 0.0;
+    }
+
+    for (long long tile0 = 0; tile0 < n_src; tile0 += TILE) {
+        const int m = (int)((n_src - tile0 < TILE) ? (n_src - tile0) : TILE);
+        __syncthreads();
+        for (int i = tid; i < m * 3; i += BLOCK)
+            lds_r[i] = r_src[tile0 * 3 + i];
+        for (int i = tid; i < m * K::SRCDIM; i += BLOCK)
+            lds_f[i] = f_src[tile0 * K::SRCDIM + i];
+        __syncthreads();
+
+        for (int s = 0; s < m; ++s) {
+            double sp[3], sf[K::SRCDIM];
+#pragma unroll
+            for (int j = 0; j < 3; ++j)
+                sp[j] = lds_r[3 * s + j];
+#pragma unroll
+            for (int j = 0; j < K::SRCDIM; ++j)
+                sf[j] = lds_f[K::SRCDIM * s + j];
+#pragma unroll
+            for (int k = 0; k < TPT; ++k)
+                K::pair(tp[k], sp, sf, acc[k], params);
+        }
+    }
+
+#pragma unroll
+    for (int k = 0; k < TPT; ++k) {
+        const long long it = base + (long long)k * BLOCK + tid;
+        if (it < n_trg) {
+            u_trg[3 * it + 0] = K::finish(acc[k][0], params);
+            u_trg[3 * it + 1] = K::finish(acc[k][1], params);
+            u_trg[3 * it + 2] = K::finish(acc[k][2], params);
+        }
+    }
+}
+
+/* ---- launch helpers (host) -------------------------------------------- */
+
+namespace skelly {
+
+static inline int pick_tpt(long long n_trg) {
+    /* Keep >= ~400 workgroups in flight (256 CUs, 8 XCDs) before widening
+     * per-thread work. */
+    for (int tpt : {4, 2}) {
+        long long blocks = (n_trg + (long long)BLOCK * tpt - 1) / ((long long)BLOCK * tpt);
+        if (blocks >= 400)
+            return tpt;
+    }
+    return 1;
+}
+
+template <typename K>
+hipError_t launch_pair(const double *r_src, const double *f_src, const double *r_trg,
+                       double *u_trg, long long n_src, long long n_trg,
+                       typename K::Params params, hipStream_t stream) {
+    if (n_trg <= 0)
+        return hipSuccess;
+    const int tpt = pick_tpt(n_trg);
+    const long long blocks = (n_trg + (long long)BLOCK * tpt - 1) / ((long long)BLOCK * tpt);
+    dim3 grid((unsigned)blocks), block(BLOCK);
+    switch (tpt) {
+    case 4:
+        hipLaunchKernelGGL((pair_driver<K, 4>), grid, block, 0, stream, r_src, f_src, r_trg,
+                           u_trg, n_src, n_trg, params);
+        break;
+    case 2:
+        hipLaunchKernelGGL((pair_driver<K, 2>), grid, block, 0, stream, r_src, f_src, r_trg,
+                           u_trg, n_src, n_trg, params);
+        break;
+    default:
+        hipLaunchKernelGGL((pair_driver<K, 1>), grid, block, 0, stream, r_src, f_src, r_trg,
+                           u_trg, n_src, n_trg, params);
+        break;
+    }
+    return hipGetLastError();
+}
+
+hipError_t launch_stokeslet(const double *r_src, const double *f_src, const double *r_trg,
+                            double *u_trg, long long n_src, long long n_trg, double scale,
+                            hipStream_t stream) {
+    Stokeslet::Params p{scale};
+    return launch_pair<Stokeslet>(r_src, f_src, r_trg, u_trg, n_src, n_trg, p, stream);
+}
+
+hipError_t launch_stresslet(const double *r_src, const double *f_src, const double *r_trg,
+                            double *u_trg, long long n_src, long long n_trg, double scale,
+                            hipStream_t stream) {
+    Stresslet::Params p{scale};
+    return launch_pair<Stresslet>(r_src, f_src, r_trg, u_trg, n_src, n_trg, p, stream);
+}
+
+hipError_t launch_oseen(const double *r_src, const double *density, const double *r_trg,
+                        double *u_trg, long long n_src, long long n_trg, double factor,
+                        double reg, double eps, hipStream_t stream) {
+    OseenContract::Params p{factor, reg * reg, eps * eps};
+    return launch_pair<OseenContract>(r_src, density, r_trg, u_trg, n_src, n_trg, p, stream);
+}
+
+hipError_t launch_rotlet(const double *r_src, const double *density, const double *r_trg,
+                         double *u_trg, long long n_src, long long n_trg, double factor,
+                         double reg, double eps, hipStream_t stream) {
+    Rotlet::Params p{factor, reg * reg, eps * eps};
+    return launch_pair<Rotlet>(r_src, density, r_trg, u_trg, n_src, n_trg, p, stream);
+}
+
+/* ---- fp64 FMA peak microbenchmark (roofline denominator) -------------- */
+
+__global__ __launch_bounds__(256) void fp64_fma_peak_kernel(double *out, int iters) {
+    double a0 = 1.0 + 1e-9 * threadIdx.x, a1 = a0 + 0.1, a2 = a0 + 0.2, a3 = a0 + 0.3;
+    double a4 = a0 + 0.4, a5 = a0 + 0.5, a6 = a0 + 0.6, a7 = a0 + 0.7;
+    const double b = 1.0 + 1e-12, c = 1e-12;
+    for (int i = 0; i < iters; ++i) {
+        a0 = __builtin_fma(a0, b, c);
+        a1 = __builtin_fma(a1, b, c);
+        a2 = __builtin_fma(a2, b, c);
+        a3 = __builtin_fma(a3, b, c);
+        a4 = __builtin_fma(a4, b, c);
+        a5 = __builtin_fma(a5, b, c);
+        a6 = __builtin_fma(a6, b, c);
+        a7 = __builtin_fma(a7, b, c);
+    }
+    const double s = a0 + a1 + a2 + a3 + a4 + a5 + a6 + a7;
+    if (s == -1.0) /* never true; defeats DCE without a store per thread */
+        out[blockIdx.x] = s;
+}
+
+hipError_t run_fp64_peak(double *out_tflops) {
+    const int iters = 100000, blocks = 2048, threads = 256;
+    double *d;
+    hipError_t err = hipMalloc(&d, blocks * sizeof(double));
+    if (err != hipSuccess)
+        return err;
+    hipEvent_t t0, t1;
+    hipEventCreate(&t0);
+    hipEventCreate(&t1);
+    /* warmup */
+    hipLaunchKernelGGL(fp64_fma_peak_kernel, dim3(blocks), dim3(threads), 0, 0, d, iters / 10);
+    hipEventRecord(t0);
+    hipLaunchKernelGGL(fp64_fma_peak_kernel, dim3(blocks), dim3(threads), 0, 0, d, iters);
+    hipEventRecord(t1);
+    err = hipEventSynchronize(t1);
+    float ms = 0;
+    hipEventElapsedTime(&ms, t0, t1);
+    hipEventDestroy(t0);
+    hipEventDestroy(t1);
+    hipFree(d);
+    if (err != hipSuccess)
+        return err;
+    const double flops = 2.0 * 8.0 * (double)iters * (double)blocks * threads;
+    *out_tflops = flops / (ms * 1e-3) / 1e12;
+    return hipSuccess;
+}
+
+} // namespace skelly
