@@ -41,9 +41,20 @@
 __device__ inline double rsq_refined(double x) {
     double y;
     asm("v_rsq_f64 %0, %1" : "=v"(y) : "v"(x));
+#ifdef SKELLY_RSQ_NEWTON
+    /* 4-op quadratic step: y' = -(0.5*y) * ((x*y)*y - 3). With v_rsq_f64's
+     * ~2^-26 seed this lands at ~1e-14 relative per pair — well inside the
+     * 1e-10 parity bar — and saves one fp64 slot per pair. */
+    const double t = x * y;
+    const double w = __builtin_fma(t, y, -3.0);
+    return (-0.5 * y) * w;
+#else
+    /* 5-op Householder (2nd order) step — the same polish ROCm libm applies:
+     * cubic convergence, <= 2 ulp of fp64. */
     const double e = __builtin_fma(-x * y, y, 1.0);
     const double c = __builtin_fma(e, 0.375, 0.5);
     return __builtin_fma(y * e, c, y);
+#endif
 }
 
 /* ---- kernel functors ------------------------------------------------- */
