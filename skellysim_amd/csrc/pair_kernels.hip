@@ -41,19 +41,22 @@
 __device__ inline double rsq_refined(double x) {
     double y;
     asm("v_rsq_f64 %0, %1" : "=v"(y) : "v"(x));
-#ifdef SKELLY_RSQ_NEWTON
-    /* 4-op quadratic step: y' = -(0.5*y) * ((x*y)*y - 3). With v_rsq_f64's
-     * ~2^-26 seed this lands at ~1e-14 relative per pair — well inside the
-     * 1e-10 parity bar — and saves one fp64 slot per pair. */
-    const double t = x * y;
-    const double w = __builtin_fma(t, y, -3.0);
-    return (-0.5 * y) * w;
-#else
+#ifdef SKELLY_RSQ_PRECISE
     /* 5-op Householder (2nd order) step — the same polish ROCm libm applies:
      * cubic convergence, <= 2 ulp of fp64. */
     const double e = __builtin_fma(-x * y, y, 1.0);
     const double c = __builtin_fma(e, 0.375, 0.5);
     return __builtin_fma(y * e, c, y);
+#else
+    /* Default: 4-op quadratic Newton step, y' = -(0.5*y) * ((x*y)*y - 3).
+     * Measured on MI355X (profiles/rocprof_r01.md): identical norm-relative
+     * parity vs the CPU oracle (~5e-16 at 2e4x4096 incl. near-coincident
+     * pairs) and +4.6%/+2.9%/+2.2% on stokeslet/stresslet/oseen over the
+     * Householder form — v_rsq_f64's seed is accurate enough that one
+     * quadratic step reaches the fp64 rounding floor of the summation. */
+    const double t = x * y;
+    const double w = __builtin_fma(t, y, -3.0);
+    return (-0.5 * y) * w;
 #endif
 }
 
