@@ -45,24 +45,31 @@ def parse_args():
                    help="cloud size (sources == targets); default the metric point 1e6")
     p.add_argument("--seed", type=int, default=100)
     p.add_argument("--skip-cpu-baseline", action="store_true")
+    p.add_argument("--kernel", default="stokeslet", choices=["stokeslet", "oseen"],
+                   help="stokeslet = the headline metric; oseen = BASELINE config 3 "
+                        "(regularized Stokeslet)")
     return p.parse_args()
 
 
-def cpu_baseline(r_src, f_src, r_trg, eta, budget_s=15.0):
+def cpu_baseline(r_src, f_src, r_trg, eta, kernel="stokeslet", budget_s=15.0):
     """Time the C oracle (reference CPU path restated; kind='port') on a
     bounded target sample of the same workload; returns pairs/s."""
     import oracle
+    if kernel == "stokeslet":
+        fn = lambda t: oracle.stokeslet(r_src, f_src, t, eta)
+    else:
+        fn = lambda t: oracle.oseen_contract(r_src, t, f_src, eta)
     n_src = len(r_src)
     cores = oracle.num_threads()
     # probe to pick a sample that costs ~budget_s
     probe_t = min(256, len(r_trg))
     t0 = time.perf_counter()
-    oracle.stokeslet(r_src, f_src, r_trg[:probe_t], eta)
+    fn(r_trg[:probe_t])
     dt = time.perf_counter() - t0
     rate = probe_t * n_src / max(dt, 1e-9)
     sample_t = int(min(len(r_trg), max(probe_t, rate * budget_s / n_src)))
     t0 = time.perf_counter()
-    oracle.stokeslet(r_src, f_src, r_trg[:sample_t], eta)
+    fn(r_trg[:sample_t])
     dt = time.perf_counter() - t0
     return {
         "value": sample_t * n_src / dt,
@@ -96,6 +103,13 @@ def main():
     if not torch.cuda.is_available():
         print(json.dumps({"error": "no GPU visible; bench.py must run on an MI355X box"}))
         sys.exit(1)
+
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        # launched directly with --gpus N: re-exec under torchrun
+        os.execvp(sys.executable, [
+            sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+            f"--nproc-per-node={args.gpus}", "--master-addr", "127.0.0.1",
+            "--master-port", "29571", os.path.abspath(__file__)] + sys.argv[1:])
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -131,6 +145,13 @@ def main():
     ev_start = [torch.cuda.Event(enable_timing=True) for _ in range(args.steps)]
     ev_end = [torch.cuda.Event(enable_timing=True) for _ in range(args.steps)]
 
+    if args.kernel == "stokeslet":
+        kernel_fn = lambda r, f, t, out: ska.stokeslet_device(r, f, t, eta, out=out)
+        flops_per_pair, metric = FLOPS_PER_PAIR, "Stokeslet pair-interactions/sec (fp64)"
+    else:  # regularized Stokeslet (BASELINE config 3); defaults kernels.hpp:34-35
+        kernel_fn = lambda r, f, t, out: ska.oseen_contract_device(r, t, f, eta, out=out)
+        flops_per_pair, metric = 32.0, "Regularized-Stokeslet pair-interactions/sec (fp64)"
+
     def step(i=None):
         if world > 1:
             r_all = allgather_rows(r_src_local)
@@ -139,7 +160,7 @@ def main():
             r_all, f_all = r_src_local, f_src_local
         if i is not None:
             ev_start[i].record()
-        ska.stokeslet_device(r_all, f_all, r_trg_local, eta, out=u_local)
+        kernel_fn(r_all, f_all, r_trg_local, u_local)
         if i is not None:
             ev_end[i].record()
 
@@ -173,8 +194,8 @@ def main():
         kernel_ms = [s.elapsed_time(t) for s, t in zip(ev_start, ev_end)]
         kernel_s = float(np.mean(kernel_ms)) / 1e3
         pairs_per_launch = float(n) * (te - ts)
-        achieved_tflops = FLOPS_PER_PAIR * pairs_per_launch / kernel_s / 1e12
-        traffic = load_traffic_calibration(n, te - ts)
+        achieved_tflops = flops_per_pair * pairs_per_launch / kernel_s / 1e12
+        traffic = load_traffic_calibration(n, te - ts) if args.kernel == "stokeslet" else None
 
         peak_meas = ctypes.c_double(0.0)
         try:
@@ -184,10 +205,10 @@ def main():
 
         cpu = None
         if world == 1 and not args.skip_cpu_baseline:
-            cpu = cpu_baseline(pts, strengths, pts, eta)
+            cpu = cpu_baseline(pts, strengths, pts, eta, kernel=args.kernel)
 
         out = {
-            "metric": "Stokeslet pair-interactions/sec (fp64)",
+            "metric": metric,
             "value": pairs_per_s,
             "unit": "pairs/s",
             "n_gpus": world,
@@ -200,8 +221,8 @@ def main():
             "dtype": "f64",
             "data": "synthetic",
             "config": {
-                "workload": "stokeslet_direct_N1e6" if n == 1_000_000
-                            else f"stokeslet_direct_N{n}",
+                "workload": f"{args.kernel}_direct_N1e6" if n == 1_000_000
+                            else f"{args.kernel}_direct_N{n}",
                 "n_src": n,
                 "n_trg": n,
                 "seed": args.seed,
