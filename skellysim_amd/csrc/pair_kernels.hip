@@ -15,13 +15,15 @@
  *   - Sources staged through LDS in TILE=512-point tiles (24 KB stokeslet,
  *     48 KB stresslet). Inner-loop LDS reads are wave-uniform -> broadcast,
  *     conflict-free by construction.
- *   - fp64 rsqrt: raw v_rsq_f64 + one 5-op Householder step (the same
- *     refinement ROCm's libm applies after v_rsq_f64), skipping libm's
- *     fp-class edge handling because r^2 is a finite sum of squares and the
- *     r^2==0 case is masked explicitly (reference masks it too,
- *     kernels.cu:39,70).
+ *   - fp64 rsqrt: raw v_rsq_f64 + one 4-op Newton step (see rsq_refined),
+ *     skipping libm's fp-class edge handling because r^2 is a finite sum of
+ *     squares and the r^2==0 case is masked explicitly (reference masks it
+ *     too, kernels.cu:39,70).
  *   - Accumulation order per target is source order (tile-major), fixed ->
- *     bit-reproducible for a given shard layout.
+ *     bit-reproducible for a given shard layout and slice count.
+ *   - Small target counts (< 512 workgroups at one target/thread) switch to
+ *     a source-split launch: gridDim.y source slices accumulate partials
+ *     that a second kernel reduces in fixed slice order (deterministic).
  *
  * Roofline: compute-bound on the fp64 VALU. The kernels are
  * deliberately not GEMM-shaped: gfx950's fp64 matrix rate equals its vector
@@ -33,11 +35,8 @@
 #define BLOCK 256
 #define TILE 512
 
-/* v_rsq_f64 + one Householder(2nd order) refinement.
- * v_rsq_f64 gives ~2^-26 relative error; the cubic-convergence step below
- * (e = 1 - x*y^2; y += y*e*(0.5 + 0.375*e)) lands at <=2 ulp of fp64, which
- * is the same polish ROCm libm's rsqrt(double) performs. Inputs here are
- * finite and >= 0; x == 0 must be masked by the caller. */
+/* v_rsq_f64 + one refinement step. Inputs here are finite and >= 0;
+ * x == 0 must be masked by the caller. */
 __device__ inline double rsq_refined(double x) {
     double y;
     asm("v_rsq_f64 %0, %1" : "=v"(y) : "v"(x));
@@ -196,7 +195,13 @@ struct Rotlet {
 
 /* ---- driver ----------------------------------------------------------- */
 
-template <typename K, int TPT>
+/* PARTIAL=false: each block sums ALL sources for its targets and writes the
+ * finished (scaled) velocities. PARTIAL=true (source-split, used when small
+ * target counts would underfill the 256 CUs): gridDim.y slices partition the
+ * sources; block (x, y) sums slice y for target tile x into
+ * u_trg[y * 3*n_trg ...] UNSCALED; reduce_partials then sums the slices in
+ * fixed slice order (deterministic) and applies the finish scale. */
+template <typename K, int TPT, bool PARTIAL = false>
 __global__ __launch_bounds__(BLOCK) void pair_driver(const double *__restrict__ r_src,
                                                      const double *__restrict__ f_src,
                                                      const double *__restrict__ r_trg,
@@ -221,8 +226,15 @@ __global__ __launch_bounds__(BLOCK) void pair_driver(const double *__restrict__ 
         acc[k][0] = acc[k][1] = acc[k][2] = 0.0;
     }
 
-    for (long long tile0 = 0; tile0 < n_src; tile0 += TILE) {
-        const int m = (int)((n_src - tile0 < TILE) ? (n_src - tile0) : TILE);
+    long long src_begin = 0, src_end = n_src;
+    if (PARTIAL) {
+        const long long chunk = (n_src + gridDim.y - 1) / gridDim.y;
+        src_begin = (long long)blockIdx.y * chunk;
+        src_end = src_begin + chunk < n_src ? src_begin + chunk : n_src;
+    }
+
+    for (long long tile0 = src_begin; tile0 < src_end; tile0 += TILE) {
+        const int m = (int)((src_end - tile0 < TILE) ? (src_end - tile0) : TILE);
         __syncthreads();
         for (int i = tid; i < m * 3; i += BLOCK)
             lds_r[i] = r_src[tile0 * 3 + i];
@@ -271,11 +283,34 @@ __global__ __launch_bounds__(BLOCK) void pair_driver(const double *__restrict__ 
     for (int k = 0; k < TPT; ++k) {
         const long long it = base + (long long)k * BLOCK + tid;
         if (it < n_trg) {
-            u_trg[3 * it + 0] = K::finish(acc[k][0], params);
-            u_trg[3 * it + 1] = K::finish(acc[k][1], params);
-            u_trg[3 * it + 2] = K::finish(acc[k][2], params);
+            if (PARTIAL) {
+                double *p = u_trg + (long long)blockIdx.y * 3 * n_trg;
+                p[3 * it + 0] = acc[k][0];
+                p[3 * it + 1] = acc[k][1];
+                p[3 * it + 2] = acc[k][2];
+            } else {
+                u_trg[3 * it + 0] = K::finish(acc[k][0], params);
+                u_trg[3 * it + 1] = K::finish(acc[k][1], params);
+                u_trg[3 * it + 2] = K::finish(acc[k][2], params);
+            }
         }
     }
+}
+
+/* Sum source-slice partials in fixed slice order and apply the finish scale.
+ * partial layout: [n_slices][n_trg][3]; one thread per (target, component). */
+template <typename K>
+__global__ __launch_bounds__(BLOCK) void reduce_partials(const double *__restrict__ partial,
+                                                         double *__restrict__ u_trg,
+                                                         long long n_trg, int n_slices,
+                                                         typename K::Params params) {
+    const long long i = (long long)blockIdx.x * BLOCK + threadIdx.x; /* 3*n_trg elems */
+    if (i >= 3 * n_trg)
+        return;
+    double s = 0.0;
+    for (int y = 0; y < n_slices; ++y)
+        s += partial[(long long)y * 3 * n_trg + i];
+    u_trg[i] = K::finish(s, params);
 }
 
 /* ---- launch helpers (host) -------------------------------------------- */
@@ -299,24 +334,56 @@ hipError_t launch_pair(const double *r_src, const double *f_src, const double *r
                        typename K::Params params, hipStream_t stream) {
     if (n_trg <= 0)
         return hipSuccess;
-    const int tpt = pick_tpt(n_trg);
-    const long long blocks = (n_trg + (long long)BLOCK * tpt - 1) / ((long long)BLOCK * tpt);
-    dim3 grid((unsigned)blocks), block(BLOCK);
-    switch (tpt) {
-    case 4:
-        hipLaunchKernelGGL((pair_driver<K, 4>), grid, block, 0, stream, r_src, f_src, r_trg,
-                           u_trg, n_src, n_trg, params);
-        break;
-    case 2:
-        hipLaunchKernelGGL((pair_driver<K, 2>), grid, block, 0, stream, r_src, f_src, r_trg,
-                           u_trg, n_src, n_trg, params);
-        break;
-    default:
-        hipLaunchKernelGGL((pair_driver<K, 1>), grid, block, 0, stream, r_src, f_src, r_trg,
-                           u_trg, n_src, n_trg, params);
-        break;
+    /* Source-split when the target grid alone underfills the chip (256 CUs;
+     * aim >= ~512 workgroups) and there are enough sources to slice. */
+    const long long blocks_t1 = (n_trg + BLOCK - 1) / BLOCK;
+    int n_slices = 1;
+    if (blocks_t1 < 512) {
+        long long s = (512 + blocks_t1 - 1) / blocks_t1;
+        long long max_by_src = (n_src + 2047) / 2048; /* keep >= ~2048 src/slice */
+        n_slices = (int)(s < max_by_src ? s : max_by_src);
+        if (n_slices < 1)
+            n_slices = 1;
     }
-    return hipGetLastError();
+
+    dim3 block(BLOCK);
+    if (n_slices == 1) {
+        const int tpt = pick_tpt(n_trg);
+        const long long blocks =
+            (n_trg + (long long)BLOCK * tpt - 1) / ((long long)BLOCK * tpt);
+        dim3 grid((unsigned)blocks);
+        switch (tpt) {
+        case 4:
+            hipLaunchKernelGGL((pair_driver<K, 4>), grid, block, 0, stream, r_src, f_src,
+                               r_trg, u_trg, n_src, n_trg, params);
+            break;
+        case 2:
+            hipLaunchKernelGGL((pair_driver<K, 2>), grid, block, 0, stream, r_src, f_src,
+                               r_trg, u_trg, n_src, n_trg, params);
+            break;
+        default:
+            hipLaunchKernelGGL((pair_driver<K, 1>), grid, block, 0, stream, r_src, f_src,
+                               r_trg, u_trg, n_src, n_trg, params);
+            break;
+        }
+        return hipGetLastError();
+    }
+
+    /* split path: one target per thread (TPT=1), gridDim.y source slices */
+    double *workspace = nullptr;
+    const size_t ws_bytes = (size_t)n_slices * 3 * n_trg * sizeof(double);
+    hipError_t err = hipMallocAsync((void **)&workspace, ws_bytes, stream);
+    if (err != hipSuccess)
+        return err;
+    dim3 grid((unsigned)blocks_t1, (unsigned)n_slices);
+    hipLaunchKernelGGL((pair_driver<K, 1, true>), grid, block, 0, stream, r_src, f_src, r_trg,
+                       workspace, n_src, n_trg, params);
+    const long long rblocks = (3 * n_trg + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL((reduce_partials<K>), dim3((unsigned)rblocks), block, 0, stream,
+                       workspace, u_trg, n_trg, n_slices, params);
+    err = hipGetLastError();
+    hipError_t err2 = hipFreeAsync(workspace, stream);
+    return err != hipSuccess ? err : err2;
 }
 
 hipError_t launch_stokeslet(const double *r_src, const double *f_src, const double *r_trg,
@@ -375,19 +442,19 @@ hipError_t run_fp64_peak(double *out_tflops) {
     if (err != hipSuccess)
         return err;
     hipEvent_t t0, t1;
-    hipEventCreate(&t0);
-    hipEventCreate(&t1);
+    (void)hipEventCreate(&t0);
+    (void)hipEventCreate(&t1);
     /* warmup */
     hipLaunchKernelGGL(fp64_fma_peak_kernel, dim3(blocks), dim3(threads), 0, 0, d, iters / 10);
-    hipEventRecord(t0);
+    (void)hipEventRecord(t0);
     hipLaunchKernelGGL(fp64_fma_peak_kernel, dim3(blocks), dim3(threads), 0, 0, d, iters);
-    hipEventRecord(t1);
+    (void)hipEventRecord(t1);
     err = hipEventSynchronize(t1);
     float ms = 0;
-    hipEventElapsedTime(&ms, t0, t1);
-    hipEventDestroy(t0);
-    hipEventDestroy(t1);
-    hipFree(d);
+    (void)hipEventElapsedTime(&ms, t0, t1);
+    (void)hipEventDestroy(t0);
+    (void)hipEventDestroy(t1);
+    (void)hipFree(d);
     if (err != hipSuccess)
         return err;
     const double flops = 2.0 * 8.0 * (double)iters * (double)blocks * threads;

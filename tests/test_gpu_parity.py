@@ -108,6 +108,29 @@ def test_stresslet_sizes_vs_oracle(ska, oracle_mod, n_src, n_trg):
     assert rel(u, ref) < REL_TOL
 
 
+@pytest.mark.parametrize("n_src,n_trg", [
+    (20000, 8192),    # split path: 32 target blocks -> ~10 source slices
+    (100000, 4096),   # deeper split
+    (5000, 100),      # tiny targets, few slices
+])
+def test_source_split_path_vs_oracle(ska, oracle_mod, n_src, n_trg):
+    """Small-target launches take the source-split (split-K) path; results
+    must still match the oracle and be deterministic."""
+    rng = np.random.default_rng(n_src ^ n_trg)
+    r_src = rng.uniform(-1, 1, (n_src, 3))
+    f_src = rng.uniform(-1, 1, (n_src, 3))
+    r_trg = rng.uniform(-1, 1, (n_trg, 3))
+    u = ska.stokeslet_direct_gpu(r_src, None, r_trg, f_src, None, 1.3)
+    ref = oracle_mod.stokeslet(r_src, f_src, r_trg, 1.3)
+    assert rel(u, ref) < REL_TOL
+    u2 = ska.stokeslet_direct_gpu(r_src, None, r_trg, f_src, None, 1.3)
+    assert np.array_equal(u, u2)
+    f9 = rng.uniform(-1, 1, (n_src, 9))
+    u9 = ska.stresslet_direct_gpu(None, r_src, r_trg, None, f9, 1.3)
+    ref9 = oracle_mod.stresslet(r_src, f9, r_trg, 1.3)
+    assert rel(u9, ref9) < REL_TOL
+
+
 def test_empty_sources_and_targets(ska):
     r = np.random.default_rng(0).uniform(-1, 1, (16, 3))
     f = np.ones((16, 3))
