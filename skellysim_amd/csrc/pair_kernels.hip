@@ -318,13 +318,13 @@ __global__ __launch_bounds__(BLOCK) void reduce_partials(const double *__restric
 namespace skelly {
 
 static inline int pick_tpt(long long n_trg) {
-    /* Keep >= ~400 workgroups in flight (256 CUs, 8 XCDs) before widening
-     * per-thread work. */
-    for (int tpt : {4, 2}) {
-        long long blocks = (n_trg + (long long)BLOCK * tpt - 1) / ((long long)BLOCK * tpt);
-        if (blocks >= 400)
-            return tpt;
-    }
+    /* Prefer 4 targets/thread (amortizes the per-source LDS broadcast reads
+     * 4x); occupancy for small target counts is recovered by source
+     * splitting, not by shrinking TPT. */
+    if (n_trg >= BLOCK * 4)
+        return 4;
+    if (n_trg >= BLOCK * 2)
+        return 2;
     return 1;
 }
 
@@ -334,12 +334,14 @@ hipError_t launch_pair(const double *r_src, const double *f_src, const double *r
                        typename K::Params params, hipStream_t stream) {
     if (n_trg <= 0)
         return hipSuccess;
+    const int tpt = pick_tpt(n_trg);
+    const long long blocks = (n_trg + (long long)BLOCK * tpt - 1) / ((long long)BLOCK * tpt);
+
     /* Source-split when the target grid alone underfills the chip (256 CUs;
      * aim >= ~512 workgroups) and there are enough sources to slice. */
-    const long long blocks_t1 = (n_trg + BLOCK - 1) / BLOCK;
     int n_slices = 1;
-    if (blocks_t1 < 512) {
-        long long s = (512 + blocks_t1 - 1) / blocks_t1;
+    if (blocks < 512) {
+        long long s = (512 + blocks - 1) / blocks;
         long long max_by_src = (n_src + 2047) / 2048; /* keep >= ~2048 src/slice */
         n_slices = (int)(s < max_by_src ? s : max_by_src);
         if (n_slices < 1)
@@ -348,9 +350,6 @@ hipError_t launch_pair(const double *r_src, const double *f_src, const double *r
 
     dim3 block(BLOCK);
     if (n_slices == 1) {
-        const int tpt = pick_tpt(n_trg);
-        const long long blocks =
-            (n_trg + (long long)BLOCK * tpt - 1) / ((long long)BLOCK * tpt);
         dim3 grid((unsigned)blocks);
         switch (tpt) {
         case 4:
@@ -369,15 +368,27 @@ hipError_t launch_pair(const double *r_src, const double *f_src, const double *r
         return hipGetLastError();
     }
 
-    /* split path: one target per thread (TPT=1), gridDim.y source slices */
+    /* split path: same TPT, gridDim.y source slices into a partial buffer */
     double *workspace = nullptr;
     const size_t ws_bytes = (size_t)n_slices * 3 * n_trg * sizeof(double);
     hipError_t err = hipMallocAsync((void **)&workspace, ws_bytes, stream);
     if (err != hipSuccess)
         return err;
-    dim3 grid((unsigned)blocks_t1, (unsigned)n_slices);
-    hipLaunchKernelGGL((pair_driver<K, 1, true>), grid, block, 0, stream, r_src, f_src, r_trg,
-                       workspace, n_src, n_trg, params);
+    dim3 grid((unsigned)blocks, (unsigned)n_slices);
+    switch (tpt) {
+    case 4:
+        hipLaunchKernelGGL((pair_driver<K, 4, true>), grid, block, 0, stream, r_src, f_src,
+                           r_trg, workspace, n_src, n_trg, params);
+        break;
+    case 2:
+        hipLaunchKernelGGL((pair_driver<K, 2, true>), grid, block, 0, stream, r_src, f_src,
+                           r_trg, workspace, n_src, n_trg, params);
+        break;
+    default:
+        hipLaunchKernelGGL((pair_driver<K, 1, true>), grid, block, 0, stream, r_src, f_src,
+                           r_trg, workspace, n_src, n_trg, params);
+        break;
+    }
     const long long rblocks = (3 * n_trg + BLOCK - 1) / BLOCK;
     hipLaunchKernelGGL((reduce_partials<K>), dim3((unsigned)rblocks), block, 0, stream,
                        workspace, u_trg, n_trg, n_slices, params);
