@@ -63,6 +63,58 @@ def _worker(rank, world, init_file, n_src, n_trg, q):
         dist.destroy_process_group()
 
 
+def _shell_worker(rank, world, init_file, N, q):
+    import torch.distributed as dist
+    from skellysim_amd.flows import ShellOperator
+    from skellysim_amd.sharded import shard_range
+
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    try:
+        rng = np.random.default_rng(77)
+        M = rng.uniform(-1, 1, (3 * N, 3 * N))
+        x = rng.uniform(-1, 1, 3 * N)
+        v = rng.uniform(-1, 1, 3 * N)
+        a, b = shard_range(3 * N, world, rank)  # row-block distribution
+        op = ShellOperator(torch.from_numpy(M[a:b].copy()),
+                           torch.from_numpy(M[a:b].copy()), distributed=True)
+        p = op.apply_preconditioner(torch.from_numpy(x[a:b].copy()))
+        m = op.matvec(torch.from_numpy(x[a:b].copy()), torch.from_numpy(v[a:b].copy()))
+        q.put((rank, p.numpy(), m.numpy()))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_shell_operator_row_sharded_gloo():
+    """Row-sharded periphery GEMVs over gloo (mirrors the reference's
+    Allgatherv + row-distributed dense ops, periphery.cpp:21-47,422-442)."""
+    N = 37  # 3N = 111 rows, uneven split across 2 ranks
+    with tempfile.TemporaryDirectory() as td:
+        init_file = os.path.join(td, "pg2")
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        procs = [ctx.Process(target=_shell_worker, args=(r, WORLD, init_file, N, q))
+                 for r in range(WORLD)]
+        for p in procs:
+            p.start()
+        results = {}
+        for _ in range(WORLD):
+            rank, pre, mv = q.get(timeout=150)
+            results[rank] = (pre, mv)
+        for p in procs:
+            p.join(timeout=60)
+            assert p.exitcode == 0
+    rng = np.random.default_rng(77)
+    M = rng.uniform(-1, 1, (3 * N, 3 * N))
+    x = rng.uniform(-1, 1, 3 * N)
+    v = rng.uniform(-1, 1, 3 * N)
+    pre = np.concatenate([results[r][0] for r in range(WORLD)])
+    mv = np.concatenate([results[r][1] for r in range(WORLD)])
+    assert np.allclose(pre, M @ x, rtol=1e-13, atol=1e-13)
+    assert np.allclose(mv, M @ x + v, rtol=1e-13, atol=1e-13)
+
+
 @pytest.mark.timeout(180)
 def test_sharded_evaluator_matches_single_process_gloo():
     """world_size=2 over gloo: sharded evaluation must equal the
