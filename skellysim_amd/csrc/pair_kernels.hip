@@ -242,28 +242,30 @@ __global__ __launch_bounds__(BLOCK) void pair_driver(const double *__restrict__ 
             lds_f[i] = f_src[tile0 * K::SRCDIM + i];
         __syncthreads();
 
-        /* Unroll by 2 sources: both sources' LDS reads issue together (one
-         * lgkmcnt wait per pair of sources instead of several per source)
-         * and their v_rsq/refine chains interleave. */
+        /* Unroll by U sources: all U sources' LDS reads issue together (one
+         * lgkmcnt wait per U sources instead of several per source) and
+         * their v_rsq/refine chains interleave. */
+#ifndef SKELLY_UNROLL
+#define SKELLY_UNROLL 2
+#endif
+        constexpr int U = SKELLY_UNROLL;
         int s = 0;
-        for (; s + 2 <= m; s += 2) {
-            double sp0[3], sf0[K::SRCDIM], sp1[3], sf1[K::SRCDIM];
+        for (; s + U <= m; s += U) {
+            double sp[U][3], sf[U][K::SRCDIM];
 #pragma unroll
-            for (int j = 0; j < 3; ++j) {
-                sp0[j] = lds_r[3 * s + j];
-                sp1[j] = lds_r[3 * (s + 1) + j];
+            for (int u = 0; u < U; ++u) {
+#pragma unroll
+                for (int j = 0; j < 3; ++j)
+                    sp[u][j] = lds_r[3 * (s + u) + j];
+#pragma unroll
+                for (int j = 0; j < K::SRCDIM; ++j)
+                    sf[u][j] = lds_f[K::SRCDIM * (s + u) + j];
             }
 #pragma unroll
-            for (int j = 0; j < K::SRCDIM; ++j) {
-                sf0[j] = lds_f[K::SRCDIM * s + j];
-                sf1[j] = lds_f[K::SRCDIM * (s + 1) + j];
-            }
+            for (int u = 0; u < U; ++u)
 #pragma unroll
-            for (int k = 0; k < TPT; ++k)
-                K::pair(tp[k], sp0, sf0, acc[k], params);
-#pragma unroll
-            for (int k = 0; k < TPT; ++k)
-                K::pair(tp[k], sp1, sf1, acc[k], params);
+                for (int k = 0; k < TPT; ++k)
+                    K::pair(tp[k], sp[u], sf[u], acc[k], params);
         }
         for (; s < m; ++s) {
             double sp[3], sf[K::SRCDIM];
