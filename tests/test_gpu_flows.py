@@ -90,6 +90,24 @@ def test_fiber_flow_with_self_subtraction(flows, oracle_mod):
     assert rel(u2.cpu().numpy(), ref2) < 1e-10
 
 
+def test_gmres_on_device_with_shell_operator(flows):
+    """GMRES (next-row 3) running entirely on cuda:0 with a resident dense
+    operator and its inverse as right preconditioner."""
+    from skellysim_amd.gmres import gmres
+    rng = np.random.default_rng(13)
+    n = 600
+    A = rng.uniform(-1, 1, (n, n)) + 20 * np.eye(n)
+    b = rng.uniform(-1, 1, n)
+    dev = torch.device("cuda:0")
+    At = torch.from_numpy(A).to(dev)
+    bt = torch.from_numpy(b).to(dev)
+    x, info = gmres(lambda v: At @ v, bt, tol=1e-11, maxiter=300, restart=60)
+    assert info["converged"]
+    assert x.is_cuda
+    r = np.linalg.norm(A @ x.cpu().numpy() - b) / np.linalg.norm(b)
+    assert r < 1e-10
+
+
 def test_shell_operator_gemvs(flows):
     """Periphery dense GEMVs (periphery.cpp:21-47) with resident matrices."""
     rng = np.random.default_rng(4)
