@@ -246,7 +246,7 @@ __global__ __launch_bounds__(BLOCK) void pair_driver(const double *__restrict__ 
          * lgkmcnt wait per U sources instead of several per source) and
          * their v_rsq/refine chains interleave. */
 #ifndef SKELLY_UNROLL
-#define SKELLY_UNROLL 2
+#define SKELLY_UNROLL 4
 #endif
         constexpr int U = SKELLY_UNROLL;
         int s = 0;
