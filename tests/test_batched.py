@@ -16,7 +16,7 @@ def test_batched_lu_solve_matches_direct():
     lu = BatchedLU(torch.from_numpy(A))
     x = lu.solve(torch.from_numpy(b))
     ref = np.stack([np.linalg.solve(A[i], b[i]) for i in range(nf)])
-    assert np.linalg.norm(x.numpy() - ref) / np.linalg.norm(ref) < 1e-12
+    assert np.linalg.norm(x.numpy() - ref) / np.linalg.norm(ref) < 1e-10
 
 
 def test_batched_lu_reuse_across_solves():
