@@ -80,6 +80,46 @@ def fiber_flow(r_src, fib_forces, weights, r_trg, eta, fiber_sizes=None,
     return vel
 
 
+def body_flow(node_pos, node_normals, densities, centers, forces, torques, r_trg, eta,
+              reg=5e-3, epsilon_distance=1e-5):
+    """Velocity at r_trg due to rigid bodies (BodyContainer::flow_spherical /
+    _ellipsoidal, src/core/body_container.cpp:269-410): a stresslet from the
+    body quadrature nodes with f_dl = 2*eta*n_i*d_j (lines 299-302), plus a
+    stokeslet from the body centers with the link forces (line 327), plus a
+    rotlet from the centers with the torques (line 335; always direct)."""
+    v = periphery_flow(node_pos, node_normals, densities, r_trg, eta)
+    if centers.shape[0]:
+        v = v + stokeslet_device(centers, forces, r_trg, eta)
+        from .evaluator import rotlet_device
+        v = v + rotlet_device(centers, r_trg, torques, eta, reg, epsilon_distance)
+    return v
+
+
+def velocity_at_targets(r_trg, eta, fiber=None, shell=None, bodies=None):
+    """Composite free-space velocity at arbitrary targets — the kernel
+    portion of System::velocity_at_targets (src/core/system.cpp:330-384,
+    lines 355-359: fiber flow + body flow + shell flow summed; the reference
+    additionally overwrites points inside bodies with rigid-body velocity,
+    which is host-side post-processing outside this engine).
+
+    fiber  = dict(r_src, forces, weights[, fiber_sizes, self_stokeslets])
+    shell  = dict(node_pos, node_normal, density)
+    bodies = dict(node_pos, node_normals, densities, centers, forces, torques)
+    """
+    u = torch.zeros_like(r_trg)
+    if fiber is not None:
+        u = u + fiber_flow(fiber["r_src"], fiber["forces"], fiber["weights"], r_trg, eta,
+                           fiber_sizes=fiber.get("fiber_sizes"),
+                           self_stokeslets=fiber.get("self_stokeslets"))
+    if shell is not None:
+        u = u + periphery_flow(shell["node_pos"], shell["node_normal"], shell["density"],
+                               r_trg, eta)
+    if bodies is not None:
+        u = u + body_flow(bodies["node_pos"], bodies["node_normals"], bodies["densities"],
+                          bodies["centers"], bodies["forces"], bodies["torques"], r_trg, eta)
+    return u
+
+
 class ShellOperator:
     """HBM-resident periphery dense operators (M_inv, stresslet+complementary).
 

@@ -90,6 +90,54 @@ def test_fiber_flow_with_self_subtraction(flows, oracle_mod):
     assert rel(u2.cpu().numpy(), ref2) < 1e-10
 
 
+def test_body_flow_and_velocity_at_targets(flows, oracle_mod):
+    """BodyContainer::flow composite (stresslet nodes + stokeslet centers +
+    rotlet centers) and the velocity_at_targets sum, vs oracle composition."""
+    rng = np.random.default_rng(33)
+    eta = 1.25
+    nb_nodes, nb, t = 600, 4, 800
+    body_nodes = rng.uniform(-1, 1, (nb_nodes, 3))
+    body_norms = rng.uniform(-1, 1, (nb_nodes, 3))
+    dens = rng.uniform(-1, 1, (nb_nodes, 3))
+    centers = rng.uniform(-1, 1, (nb, 3))
+    forces = rng.uniform(-1, 1, (nb, 3))
+    torques = rng.uniform(-1, 1, (nb, 3))
+    trg = rng.uniform(-1, 1, (t, 3))
+    dev = torch.device("cuda:0")
+    T = lambda a: torch.from_numpy(a).to(dev)
+
+    u = flows.body_flow(T(body_nodes), T(body_norms), T(dens), T(centers), T(forces),
+                        T(torques), T(trg), eta)
+    torch.cuda.synchronize()
+    f_dl = 2.0 * eta * np.einsum("ni,nj->nij", body_norms, dens).reshape(-1, 9)
+    ref = (oracle_mod.stresslet(body_nodes, f_dl, trg, eta)
+           + oracle_mod.stokeslet(centers, forces, trg, eta)
+           + oracle_mod.rotlet(centers, trg, torques, eta))
+    assert rel(u.cpu().numpy(), ref) < 1e-10
+
+    # composite velocity_at_targets: fibers + shell + bodies
+    nf_nodes = 512
+    fib_pos = rng.uniform(-1, 1, (nf_nodes, 3))
+    fib_f = rng.uniform(-1, 1, (nf_nodes, 3))
+    fib_w = rng.uniform(0.1, 1, nf_nodes)
+    sh_pos = rng.uniform(-1, 1, (700, 3))
+    sh_n = rng.uniform(-1, 1, (700, 3))
+    sh_d = rng.uniform(-1, 1, (700, 3))
+    u2 = flows.velocity_at_targets(
+        T(trg), eta,
+        fiber=dict(r_src=T(fib_pos), forces=T(fib_f), weights=T(fib_w)),
+        shell=dict(node_pos=T(sh_pos), node_normal=T(sh_n), density=T(sh_d)),
+        bodies=dict(node_pos=T(body_nodes), node_normals=T(body_norms),
+                    densities=T(dens), centers=T(centers), forces=T(forces),
+                    torques=T(torques)))
+    torch.cuda.synchronize()
+    f_dl_sh = 2.0 * eta * np.einsum("ni,nj->nij", sh_n, sh_d).reshape(-1, 9)
+    ref2 = (ref
+            + oracle_mod.stokeslet(fib_pos, fib_f * fib_w[:, None], trg, eta)
+            + oracle_mod.stresslet(sh_pos, f_dl_sh, trg, eta))
+    assert rel(u2.cpu().numpy(), ref2) < 1e-10
+
+
 def test_batched_lu_on_device(flows):
     """next-row 2: batched per-fiber LU (rocSOLVER) resident on device."""
     from skellysim_amd.batched import BatchedLU
