@@ -71,12 +71,20 @@ class ShardedPairEvaluator:
     oracle-backed compute_fn to exercise the collective logic on CPU.
     """
 
-    def __init__(self, compute_fn=None, group=None):
+    def __init__(self, compute_fn=None, kernel="stokeslet", group=None):
         if compute_fn is None:
-            from .evaluator import stokeslet_device
+            from . import evaluator as ev
 
-            def compute_fn(r_all, f_all, r_trg, eta):
-                return stokeslet_device(r_all, f_all, r_trg, eta)
+            if kernel == "stokeslet":
+                compute_fn = lambda r, f, t, eta: ev.stokeslet_device(r, f, t, eta)
+            elif kernel == "stresslet":
+                compute_fn = lambda r, f, t, eta: ev.stresslet_device(r, f, t, eta)
+            elif kernel == "oseen":
+                compute_fn = lambda r, f, t, eta: ev.oseen_contract_device(r, t, f, eta)
+            elif kernel == "rotlet":
+                compute_fn = lambda r, f, t, eta: ev.rotlet_device(r, t, f, eta)
+            else:
+                raise ValueError(f"unknown kernel {kernel!r}")
 
         self._compute = compute_fn
         self._group = group
