@@ -109,6 +109,30 @@ int skelly_rotlet_device(const double *d_r_src, const double *d_r_trg,
                          long long n_src, long long n_trg,
                          double eta, double reg, double epsilon_distance, void *stream);
 
+/* Contraction of the stresslet with a normal and a density field over one
+ * point set (kernels::stresslet_times_normal_times_density,
+ * src/core/kernels.cpp:307-334; no eta dependence, factor -3/(4*pi)):
+ * out_i = factor * sum_{j != i} (d.rho_j)(d.n_j)/|d|^5 d, d = r_i - r_j,
+ * |d| < epsilon_distance regularized to sqrt(|d|^2 + reg^2).
+ * Used by the body/fiber operator assembly. Defaults reg=5e-3, eps=1e-5
+ * (include/kernels.hpp:50-51). */
+int skelly_stresslet_normal_density_host(const double *r_src, const double *normals,
+                                         const double *density, double *out, long long n,
+                                         double reg, double epsilon_distance);
+/* device form: nd is the interleaved (n, 6) [normal | density] array */
+int skelly_stresslet_normal_density_device(const double *d_r_src, const double *d_nd,
+                                           const double *d_r_trg, double *d_out,
+                                           long long n_src, long long n_trg,
+                                           double reg, double epsilon_distance, void *stream);
+
+/* Batched oseen_tensor_direct dense builder (kernels::oseen_tensor_direct,
+ * src/core/kernels.cpp:146-195, square/self form — the per-fiber
+ * self-stokeslet build, src/core/fiber_finite_difference.cpp:56):
+ * pts (nf, n, 3) -> G (nf, 3n, 3n) row-major; coincident blocks zero. */
+int skelly_oseen_tensor_batched_device(const double *d_pts, double *d_G, long long nf,
+                                       long long n, double eta, double reg,
+                                       double epsilon_distance, void *stream);
+
 /* ---- measurement helpers ---- */
 
 /* Measured fp64 FMA throughput (TFLOP/s) of the current device via a pure

@@ -52,6 +52,10 @@ def lib():
                                                ctypes.c_double, ctypes.c_double, ctypes.c_double]
         _lib.oracle_rotlet.argtypes = [d, d, d, d, ctypes.c_long, ctypes.c_long,
                                        ctypes.c_double, ctypes.c_double, ctypes.c_double]
+        _lib.oracle_stresslet_times_normal_times_density.argtypes = [
+            d, d, d, d, ctypes.c_long, ctypes.c_double, ctypes.c_double]
+        _lib.oracle_oseen_tensor.argtypes = [d, d, ctypes.c_long, ctypes.c_double,
+                                             ctypes.c_double, ctypes.c_double]
         _lib.oracle_num_threads.restype = ctypes.c_int
     return _lib
 
@@ -109,6 +113,28 @@ def rotlet(r_src, r_trg, density, eta=1.0, reg=5e-3, eps=1e-5):
     return u
 
 
+def stresslet_times_normal_times_density(r_src, normals, density, reg=5e-3, eps=1e-5):
+    """C-oracle restatement of kernels.cpp:307-334 (no eta; factor -3/4pi)."""
+    r_src, ps = _asbuf(r_src)
+    normals, pn = _asbuf(normals)
+    density, pd = _asbuf(density)
+    out = np.zeros((len(r_src), 3))
+    _, po = _asbuf(out)
+    lib().oracle_stresslet_times_normal_times_density(ps, pn, pd, po, len(r_src), reg, eps)
+    return out
+
+
+def oseen_tensor(pts, eta=1.0, reg=5e-3, eps=1e-5):
+    """C-oracle restatement of the square oseen_tensor_direct builder
+    (kernels.cpp:146-195): (n,3) -> (3n, 3n)."""
+    pts, pp = _asbuf(pts)
+    n = len(pts)
+    G = np.zeros((3 * n, 3 * n))
+    _, pg = _asbuf(G)
+    lib().oracle_oseen_tensor(pp, pg, n, eta, reg, eps)
+    return G
+
+
 # ---------------------------------------------------------------------------
 # numpy restatements (second independent statement; cross-checks the C oracle)
 # ---------------------------------------------------------------------------
@@ -161,6 +187,41 @@ def np_oseen_contract(r_src, r_trg, density, eta=1.0, reg=5e-3, eps=1e-5):
     ddotrho = np.einsum("tsi,si->ts", d, rho)
     u = np.einsum("ts,si->ti", fr, rho) + np.einsum("ts,tsi->ti", gr * ddotrho, d)
     return u
+
+
+def np_stresslet_times_normal_times_density(r, normals, density, reg=5e-3, eps=1e-5):
+    """Restates kernels.cpp:307-334: Sdn_i = -3/(4 pi) sum_{j!=i}
+    (d.rho_j)(d.n_j)/r^5 d, d = r_i - r_j, r<eps regularized."""
+    r = np.asarray(r, float)
+    n = len(r)
+    d = r[:, None, :] - r[None, :, :]
+    dr2 = np.einsum("ijk,ijk->ij", d, d)
+    rn = np.sqrt(dr2)
+    rn = np.where(rn < eps, np.sqrt(dr2 + reg * reg), rn)
+    with np.errstate(divide="ignore"):
+        rinv5 = 1.0 / rn ** 5
+    np.fill_diagonal(rinv5, 0.0)  # i == j skip
+    f0 = (np.einsum("ijk,jk->ij", d, density) * np.einsum("ijk,jk->ij", d, normals) * rinv5)
+    return -3.0 / (4.0 * np.pi) * np.einsum("ij,ijk->ik", f0, d)
+
+
+def np_oseen_tensor(pts, eta=1.0, reg=5e-3, eps=1e-5):
+    """Restates kernels.cpp:146-195 (square self form)."""
+    pts = np.asarray(pts, float)
+    n = len(pts)
+    factor = 1.0 / (8.0 * np.pi * eta)
+    d = pts[None, :, :] - pts[:, None, :]     # d[t, s] = src - trg
+    dr2 = np.einsum("tsk,tsk->ts", d, d)
+    dr = np.sqrt(dr2)
+    near = dr <= eps
+    denom = np.where(near, np.sqrt(dr2 + reg * reg), dr)
+    with np.errstate(divide="ignore"):
+        fr = np.where(dr2 == 0.0, 0.0, factor / denom)
+        gr = np.where(dr2 == 0.0, 0.0, factor / denom ** 3)
+    G = gr[:, :, None, None] * np.einsum("tsa,tsb->tsab", d, d)
+    G += fr[:, :, None, None] * np.eye(3)[None, None]
+    G[dr2 == 0.0] = 0.0
+    return G.transpose(0, 2, 1, 3).reshape(3 * n, 3 * n)
 
 
 def np_rotlet(r_src, r_trg, density, eta=1.0, reg=5e-3, eps=1e-5):

@@ -256,6 +256,102 @@ void oracle_rotlet(const double *r_src, const double *r_trg, const double *densi
 #endif
 }
 
+/* ---- stresslet_times_normal_times_density: kernels.cpp:307-334.
+ *      Sdn_i = -3/(4 pi) * sum_{j != i} (d.rho_j)(d.n_j)/|d|^5 d,
+ *      d = r_i - r_j; r_norm < eps -> r_norm = sqrt(r_norm^2 + reg^2)
+ *      (kernels.cpp:320-323); no eta dependence (factor kernels.cpp:311).
+ *      The i == j skip (kernels.cpp:316-317) is literal here; the HIP kernel
+ *      realizes it as a d==0 mask, identical because a coincident pair's
+ *      numerator (d.rho)(d.n) d is identically zero. */
+static void sndd_chunk(const double *r_src, const double *normals, const double *density,
+                       double *out, long n, long t0, long t1, double reg, double eps) {
+    const double factor = -3.0 / (4.0 * M_PI);
+    const double reg2 = reg * reg;
+    for (long i = t0; i < t1; ++i) {
+        const double xi = r_src[3 * i + 0], yi = r_src[3 * i + 1], zi = r_src[3 * i + 2];
+        double ax = 0.0, ay = 0.0, az = 0.0;
+        for (long j = 0; j < n; ++j) {
+            if (i == j)
+                continue; /* kernels.cpp:316-317 */
+            const double dx = xi - r_src[3 * j + 0];
+            const double dy = yi - r_src[3 * j + 1];
+            const double dz = zi - r_src[3 * j + 2];
+            const double dr2 = dx * dx + dy * dy + dz * dz;
+            double rn = sqrt(dr2);
+            if (rn < eps)
+                rn = sqrt(dr2 + reg2);
+            const double rinv5 = 1.0 / (rn * rn * rn * rn * rn);
+            const double ddrho = dx * density[3 * j + 0] + dy * density[3 * j + 1] +
+                                 dz * density[3 * j + 2];
+            const double ddn = dx * normals[3 * j + 0] + dy * normals[3 * j + 1] +
+                               dz * normals[3 * j + 2];
+            const double f0 = ddrho * ddn * rinv5;
+            ax += f0 * dx;
+            ay += f0 * dy;
+            az += f0 * dz;
+        }
+        out[3 * i + 0] = ax * factor;
+        out[3 * i + 1] = ay * factor;
+        out[3 * i + 2] = az * factor;
+    }
+}
+
+void oracle_stresslet_times_normal_times_density(const double *r_src, const double *normals,
+                                                 const double *density, double *out, long n,
+                                                 double reg, double eps) {
+#ifdef _OPENMP
+#pragma omp parallel
+    {
+        int start, size;
+        chunk_start_size(omp_get_thread_num(), omp_get_num_threads(), (int)n, &start, &size);
+        sndd_chunk(r_src, normals, density, out, n, start, start + size, reg, eps);
+    }
+#else
+    sndd_chunk(r_src, normals, density, out, n, 0, n, reg, eps);
+#endif
+}
+
+/* ---- oseen_tensor_direct dense builder: kernels.cpp:146-195 (square,
+ *      r_src == r_trg layout — its only production use, the per-fiber
+ *      self-stokeslet, fiber_finite_difference.cpp:56). G is (3n, 3n),
+ *      block (t, s) = fr I + gr d d^T with d = r_s - r_t; dr2==0 blocks
+ *      left zero (kernels.cpp:166-167). Row-major output (G is symmetric,
+ *      so the layout matches Eigen's col-major bit for bit). */
+void oracle_oseen_tensor(const double *pts, double *G, long n, double eta, double reg,
+                         double eps) {
+    const double factor = 1.0 / (8.0 * M_PI * eta);
+    const double reg2 = reg * reg;
+    const long ld = 3 * n;
+    for (long t = 0; t < n; ++t)
+        for (long s = 0; s < n; ++s) {
+            const double dx = pts[3 * s + 0] - pts[3 * t + 0];
+            const double dy = pts[3 * s + 1] - pts[3 * t + 1];
+            const double dz = pts[3 * s + 2] - pts[3 * t + 2];
+            const double dr2 = dx * dx + dy * dy + dz * dz;
+            double *blk = G + (3 * t) * ld + 3 * s;
+            if (dr2 == 0.0) {
+                for (int a = 0; a < 3; ++a)
+                    for (int b = 0; b < 3; ++b)
+                        blk[a * ld + b] = 0.0;
+                continue;
+            }
+            const double dr = sqrt(dr2);
+            double fr, gr;
+            if (dr > eps) {
+                fr = factor / dr;
+                gr = factor / (dr * dr * dr);
+            } else {
+                const double di = 1.0 / sqrt(dr2 + reg2);
+                fr = factor * di;
+                gr = factor * di * di * di;
+            }
+            const double d[3] = {dx, dy, dz};
+            for (int a = 0; a < 3; ++a)
+                for (int b = 0; b < 3; ++b)
+                    blk[a * ld + b] = (a == b ? fr : 0.0) + gr * d[a] * d[b];
+        }
+}
+
 int oracle_num_threads(void) {
 #ifdef _OPENMP
     int n = 0;

@@ -18,10 +18,13 @@ from .evaluator import (  # noqa: F401
     stresslet_direct_gpu,
     oseen_contract_direct_gpu,
     rotlet_gpu,
+    stresslet_times_normal_times_density,
     stokeslet_device,
     stresslet_device,
     oseen_contract_device,
     rotlet_device,
+    stresslet_normal_density_device,
+    oseen_tensor_batched_device,
 )
 from .sharded import ShardedPairEvaluator, shard_sizes, allgather_rows  # noqa: F401
 

@@ -42,6 +42,12 @@ def _bind(lib):
                                                  ctypes.c_double, pv]
     lib.skelly_rotlet_device.argtypes = [pv, pv, pv, pv, _LL, _LL,
                                          ctypes.c_double, ctypes.c_double, ctypes.c_double, pv]
+    lib.skelly_stresslet_normal_density_host.argtypes = [_DP, _DP, _DP, _DP, _LL,
+                                                          ctypes.c_double, ctypes.c_double]
+    lib.skelly_stresslet_normal_density_device.argtypes = [pv, pv, pv, pv, _LL, _LL,
+                                                            ctypes.c_double, ctypes.c_double, pv]
+    lib.skelly_oseen_tensor_batched_device.argtypes = [pv, pv, _LL, _LL, ctypes.c_double,
+                                                        ctypes.c_double, ctypes.c_double, pv]
     lib.skelly_fp64_peak_tflops.argtypes = [ctypes.POINTER(ctypes.c_double)]
 
 

@@ -193,6 +193,44 @@ struct Rotlet {
     __device__ static inline double finish(double a, const Params &p) { return a * p.scale; }
 };
 
+struct StressletNormalDensity {
+    static constexpr int SRCDIM = 6; /* normal[3] + density[3] per source */
+    struct Params {
+        double scale; /* -3/(4*pi), kernels.cpp:311 (no eta dependence) */
+        double reg2;
+        double eps2;
+    };
+    /* kernels::stresslet_times_normal_times_density (kernels.cpp:307-334):
+     * Sdn_i += (d.rho_j)(d.n_j)/r^5 d, d = trg - src; r_norm < eps ->
+     * sqrt(r^2+reg^2) (kernels.cpp:320-323). The reference's i==j skip is a
+     * d==0 mask here (identical: the numerator is ~d^3). */
+    __device__ static inline void pair(const double t[3], const double sp[3], const double f[6],
+                                       double acc[3], const Params &p) {
+        const double dx = t[0] - sp[0];
+        const double dy = t[1] - sp[1];
+        const double dz = t[2] - sp[2];
+        double dr2 = dx * dx;
+        dr2 = __builtin_fma(dy, dy, dr2);
+        dr2 = __builtin_fma(dz, dz, dr2);
+        const double denom2 = (dr2 < p.eps2) ? dr2 + p.reg2 : dr2;
+        double y = rsq_refined(denom2);
+        y = (dr2 == 0.0) ? 0.0 : y;
+        const double y2 = y * y;
+        const double rinv5 = y * y2 * y2;
+        double ddn = dx * f[0];
+        ddn = __builtin_fma(dy, f[1], ddn);
+        ddn = __builtin_fma(dz, f[2], ddn);
+        double ddrho = dx * f[3];
+        ddrho = __builtin_fma(dy, f[4], ddrho);
+        ddrho = __builtin_fma(dz, f[5], ddrho);
+        const double f0 = ddrho * ddn * rinv5;
+        acc[0] = __builtin_fma(f0, dx, acc[0]);
+        acc[1] = __builtin_fma(f0, dy, acc[1]);
+        acc[2] = __builtin_fma(f0, dz, acc[2]);
+    }
+    __device__ static inline double finish(double a, const Params &p) { return a * p.scale; }
+};
+
 /* ---- driver ----------------------------------------------------------- */
 
 /* PARTIAL=false: each block sums ALL sources for its targets and writes the
@@ -425,6 +463,65 @@ hipError_t launch_rotlet(const double *r_src, const double *density, const doubl
                          double reg, double eps, hipStream_t stream) {
     Rotlet::Params p{factor, reg * reg, eps * eps};
     return launch_pair<Rotlet>(r_src, density, r_trg, u_trg, n_src, n_trg, p, stream);
+}
+
+hipError_t launch_stresslet_normal_density(const double *r_src, const double *nd,
+                                           const double *r_trg, double *u_trg, long long n_src,
+                                           long long n_trg, double reg, double eps,
+                                           hipStream_t stream) {
+    StressletNormalDensity::Params p{-3.0 / (4.0 * M_PI), reg * reg, eps * eps};
+    return launch_pair<StressletNormalDensity>(r_src, nd, r_trg, u_trg, n_src, n_trg, p, stream);
+}
+
+/* ---- batched oseen_tensor_direct dense builder (kernels.cpp:146-195) ---
+ * points: (nf, n, 3); G: (nf, 3n, 3n) row-major (G is symmetric, so this is
+ * bit-identical to the reference's col-major). One thread per (fiber, trg,
+ * src) pair writes its 3x3 block. Setup-time op (per-fiber self-stokeslet,
+ * fiber_finite_difference.cpp:56), not GMRES-hot. */
+__global__ __launch_bounds__(BLOCK) void oseen_tensor_kernel(const double *__restrict__ pts,
+                                                             double *__restrict__ G,
+                                                             long long nf, long long n,
+                                                             double factor, double reg2,
+                                                             double eps2) {
+    const long long idx = (long long)blockIdx.x * BLOCK + threadIdx.x;
+    const long long total = nf * n * n;
+    if (idx >= total)
+        return;
+    const long long f = idx / (n * n);
+    const long long r = idx % (n * n);
+    const long long t = r / n, s = r % n;
+    const double *p = pts + f * n * 3;
+    const double dx = p[3 * s + 0] - p[3 * t + 0];
+    const double dy = p[3 * s + 1] - p[3 * t + 1];
+    const double dz = p[3 * s + 2] - p[3 * t + 2];
+    double dr2 = dx * dx;
+    dr2 = __builtin_fma(dy, dy, dr2);
+    dr2 = __builtin_fma(dz, dz, dr2);
+    const double denom2 = (dr2 > eps2) ? dr2 : dr2 + reg2;
+    double y = rsq_refined(denom2);
+    y = (dr2 == 0.0) ? 0.0 : y; /* dr2==0 -> whole block zero (cpp:166-167) */
+    const double fr = factor * y;
+    const double gr = fr * (y * y);
+    const long long ld = 3 * n;
+    double *blk = G + f * ld * ld + (3 * t) * ld + 3 * s;
+    const double d[3] = {dx, dy, dz};
+#pragma unroll
+    for (int a = 0; a < 3; ++a)
+#pragma unroll
+        for (int b = 0; b < 3; ++b)
+            blk[a * ld + b] = (a == b ? fr : 0.0) + gr * d[a] * d[b];
+}
+
+hipError_t launch_oseen_tensor_batched(const double *pts, double *G, long long nf, long long n,
+                                       double eta, double reg, double eps, hipStream_t stream) {
+    if (nf <= 0 || n <= 0)
+        return hipSuccess;
+    const double factor = 1.0 / (8.0 * M_PI * eta);
+    const long long total = nf * n * n;
+    const long long blocks = (total + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(oseen_tensor_kernel, dim3((unsigned)blocks), dim3(BLOCK), 0, stream, pts,
+                       G, nf, n, factor, reg * reg, eps * eps);
+    return hipGetLastError();
 }
 
 /* ---- fp64 FMA peak microbenchmark (roofline denominator) -------------- */

@@ -14,6 +14,7 @@
 #include <cmath>
 #include <cstdio>
 #include <cstring>
+#include <vector>
 #include <mutex>
 #include <string>
 
@@ -26,6 +27,11 @@ hipError_t launch_oseen(const double *, const double *, const double *, double *
                         long long, double, double, double, hipStream_t);
 hipError_t launch_rotlet(const double *, const double *, const double *, double *, long long,
                          long long, double, double, double, hipStream_t);
+hipError_t launch_stresslet_normal_density(const double *, const double *, const double *,
+                                           double *, long long, long long, double, double,
+                                           hipStream_t);
+hipError_t launch_oseen_tensor_batched(const double *, double *, long long, long long, double,
+                                       double, double, hipStream_t);
 hipError_t run_fp64_peak(double *);
 } // namespace skelly
 
@@ -285,6 +291,45 @@ int skelly_rotlet_device(const double *d_r_src, const double *d_r_trg, const dou
     CHK("skelly_rotlet_device",
         skelly::launch_rotlet(d_r_src, d_density, d_r_trg, d_u_trg, n_src, n_trg, factor, reg,
                               epsilon_distance, (hipStream_t)stream));
+    return 0;
+}
+
+int skelly_stresslet_normal_density_host(const double *r_src, const double *normals,
+                                         const double *density, double *out, long long n,
+                                         double reg, double epsilon_distance) {
+    /* interleave [normal | density] into the (n, 6) source-strength array */
+    std::vector<double> nd((size_t)n * 6);
+    for (long long i = 0; i < n; ++i) {
+        for (int k = 0; k < 3; ++k) {
+            nd[6 * i + k] = normals[3 * i + k];
+            nd[6 * i + 3 + k] = density[3 * i + k];
+        }
+    }
+    return host_eval("skelly_stresslet_normal_density_host", r_src, nd.data(), 6, n, r_src, out,
+                     n,
+                     [&](const double *rs, const double *fs, const double *rt, double *u,
+                         hipStream_t s) {
+                         return skelly::launch_stresslet_normal_density(
+                             rs, fs, rt, u, n, n, reg, epsilon_distance, s);
+                     });
+}
+
+int skelly_stresslet_normal_density_device(const double *d_r_src, const double *d_nd,
+                                           const double *d_r_trg, double *d_out, long long n_src,
+                                           long long n_trg, double reg, double epsilon_distance,
+                                           void *stream) {
+    CHK("skelly_stresslet_normal_density_device",
+        skelly::launch_stresslet_normal_density(d_r_src, d_nd, d_r_trg, d_out, n_src, n_trg, reg,
+                                                epsilon_distance, (hipStream_t)stream));
+    return 0;
+}
+
+int skelly_oseen_tensor_batched_device(const double *d_pts, double *d_G, long long nf,
+                                       long long n, double eta, double reg,
+                                       double epsilon_distance, void *stream) {
+    CHK("skelly_oseen_tensor_batched_device",
+        skelly::launch_oseen_tensor_batched(d_pts, d_G, nf, n, eta, reg, epsilon_distance,
+                                            (hipStream_t)stream));
     return 0;
 }
 

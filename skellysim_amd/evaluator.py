@@ -127,6 +127,22 @@ def set_evaluator(name):
     raise ValueError(f"unknown evaluator {name!r}")
 
 
+def stresslet_times_normal_times_density(r_src, normals, density, reg=5e-3,
+                                         epsilon_distance=1e-5):
+    """Mirrors kernels::stresslet_times_normal_times_density
+    (kernels.cpp:307-334; defaults kernels.hpp:50-51). Returns (n, 3);
+    no eta dependence (factor -3/(4 pi))."""
+    src = _rows(r_src, 3, "r_src")
+    nrm = _rows(normals, 3, "normals")
+    rho = _rows(density, 3, "density")
+    out = np.zeros((len(src), 3))
+    rc = _native.lib().skelly_stresslet_normal_density_host(
+        _ptr(src), _ptr(nrm), _ptr(rho), _ptr(out), len(src),
+        float(reg), float(epsilon_distance))
+    _native.check(rc, "stresslet_normal_density")
+    return out
+
+
 # ---------------------------------------------------------------------------
 # torch device-tensor API (async on the current torch stream)
 # ---------------------------------------------------------------------------
@@ -192,6 +208,45 @@ def oseen_contract_device(r_src, r_trg, density, eta=1.0, reg=5e-3, epsilon_dist
         ctypes.c_void_p(density.data_ptr()), ctypes.c_void_p(out.data_ptr()),
         len(r_src), len(r_trg), float(eta), float(reg), float(epsilon_distance), _stream_ptr())
     _native.check(rc, "oseen_contract_device")
+    return out
+
+
+def stresslet_normal_density_device(r_src, normals, density, r_trg=None, reg=5e-3,
+                                    epsilon_distance=1e-5, out=None):
+    """Device form of stresslet_times_normal_times_density; r_trg defaults to
+    r_src (the reference's self form)."""
+    import torch
+    r_src = _dev_rows(r_src, 3, "r_src")
+    normals = _dev_rows(normals, 3, "normals")
+    density = _dev_rows(density, 3, "density")
+    r_trg = r_src if r_trg is None else _dev_rows(r_trg, 3, "r_trg")
+    nd = torch.cat([normals, density], dim=1).contiguous()
+    if out is None:
+        out = torch.empty_like(r_trg)
+    rc = _native.lib().skelly_stresslet_normal_density_device(
+        ctypes.c_void_p(r_src.data_ptr()), ctypes.c_void_p(nd.data_ptr()),
+        ctypes.c_void_p(r_trg.data_ptr()), ctypes.c_void_p(out.data_ptr()),
+        len(r_src), len(r_trg), float(reg), float(epsilon_distance), _stream_ptr())
+    _native.check(rc, "stresslet_normal_density_device")
+    return out
+
+
+def oseen_tensor_batched_device(pts, eta=1.0, reg=5e-3, epsilon_distance=1e-5, out=None):
+    """Batched dense self-Oseen-tensor builder (kernels.cpp:146-195; the
+    per-fiber self-stokeslet, fiber_finite_difference.cpp:56).
+    pts: (nf, n, 3) CUDA fp64 -> G (nf, 3n, 3n)."""
+    import torch
+    if not (isinstance(pts, torch.Tensor) and pts.is_cuda and pts.dtype == torch.float64
+            and pts.dim() == 3 and pts.shape[2] == 3):
+        raise TypeError("pts: expected (nf, n, 3) fp64 CUDA tensor")
+    pts = pts.contiguous()
+    nf, n = pts.shape[0], pts.shape[1]
+    if out is None:
+        out = torch.empty((nf, 3 * n, 3 * n), dtype=torch.float64, device=pts.device)
+    rc = _native.lib().skelly_oseen_tensor_batched_device(
+        ctypes.c_void_p(pts.data_ptr()), ctypes.c_void_p(out.data_ptr()), nf, n,
+        float(eta), float(reg), float(epsilon_distance), _stream_ptr())
+    _native.check(rc, "oseen_tensor_batched_device")
     return out
 
 

@@ -95,6 +95,31 @@ def test_goldens_edge_selfdup(oracle_mod, golden_dir):
     assert np.all(np.isfinite(u)) and np.array_equal(u, g["u_rotlet"])
 
 
+def test_c_vs_numpy_stresslet_normal_density(oracle_mod, clouds):
+    r = clouds["r_src"]
+    n, rho = clouds["f3"], clouds["f3"][::-1]
+    c = oracle_mod.stresslet_times_normal_times_density(r, n, rho)
+    np_ = oracle_mod.np_stresslet_times_normal_times_density(r, n, rho)
+    assert rel(c, np_) < 1e-13
+
+
+def test_c_vs_numpy_oseen_tensor(oracle_mod, clouds):
+    """Dense self-Oseen builder (kernels.cpp:146-195): C vs numpy, symmetry,
+    zero diagonal blocks, and consistency with the contraction kernel
+    (G @ rho == oseen_contract for the same cloud)."""
+    r = clouds["r_src"][:64]
+    G = oracle_mod.oseen_tensor(r, 1.3)
+    Gn = oracle_mod.np_oseen_tensor(r, 1.3)
+    assert rel(G, Gn) < 1e-13
+    assert np.linalg.norm(G - G.T) < 1e-12
+    for i in range(len(r)):
+        assert np.all(G[3 * i: 3 * i + 3, 3 * i: 3 * i + 3] == 0.0)
+    rho = clouds["f3"][:64]
+    u_mat = (G @ rho.reshape(-1)).reshape(-1, 3)
+    u_contract = oracle_mod.oseen_contract(r, r, rho, 1.3)
+    assert rel(u_mat, u_contract) < 1e-13
+
+
 def test_empty_sources(oracle_mod):
     r_trg = np.random.default_rng(0).uniform(-1, 1, (10, 3))
     u = oracle_mod.stokeslet(np.empty((0, 3)), np.empty((0, 3)), r_trg, 1.0)
