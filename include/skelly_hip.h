@@ -125,6 +125,13 @@ int skelly_stresslet_normal_density_device(const double *d_r_src, const double *
                                            long long n_src, long long n_trg,
                                            double reg, double epsilon_distance, void *stream);
 
+/* Dense stresslet_times_normal builder (kernels::stresslet_times_normal,
+ * src/core/kernels.cpp:264-287; body operator assembly): pts/normals (n, 3)
+ * -> Snormal (3n, 3n) C row-major, diagonal blocks zero. Device pointers. */
+int skelly_stresslet_times_normal_device(const double *d_pts, const double *d_normals,
+                                         double *d_out, long long n, double reg,
+                                         double epsilon_distance, void *stream);
+
 /* Batched oseen_tensor_direct dense builder (kernels::oseen_tensor_direct,
  * src/core/kernels.cpp:146-195, square/self form — the per-fiber
  * self-stokeslet build, src/core/fiber_finite_difference.cpp:56):

@@ -224,6 +224,23 @@ def np_oseen_tensor(pts, eta=1.0, reg=5e-3, eps=1e-5):
     return G.transpose(0, 2, 1, 3).reshape(3 * n, 3 * n)
 
 
+def np_stresslet_times_normal(r, normals, reg=5e-3, eps=1e-5):
+    """Restates kernels.cpp:264-287: Snormal block (i,j) =
+    -3/(4 pi) (d.n_j)/r^5 d d^T, d = r_i - r_j; diagonal blocks zero."""
+    r = np.asarray(r, float)
+    n = len(r)
+    d = r[:, None, :] - r[None, :, :]
+    dr2 = np.einsum("ijk,ijk->ij", d, d)
+    rn = np.sqrt(dr2)
+    rn = np.where(rn < eps, np.sqrt(dr2 + reg * reg), rn)
+    with np.errstate(divide="ignore"):
+        rinv5 = 1.0 / rn ** 5
+    np.fill_diagonal(rinv5, 0.0)
+    c = -3.0 / (4.0 * np.pi) * np.einsum("ijk,jk->ij", d, normals) * rinv5
+    S = c[:, :, None, None] * np.einsum("ija,ijb->ijab", d, d)
+    return S.transpose(0, 2, 1, 3).reshape(3 * n, 3 * n)
+
+
 def np_rotlet(r_src, r_trg, density, eta=1.0, reg=5e-3, eps=1e-5):
     """Rotlet; restates src/core/kernels.cpp:206-242 (no dr==0 skip;
     dr2 < eps^2 regularized)."""

@@ -24,6 +24,7 @@ from .evaluator import (  # noqa: F401
     oseen_contract_device,
     rotlet_device,
     stresslet_normal_density_device,
+    stresslet_times_normal_device,
     oseen_tensor_batched_device,
 )
 from .sharded import ShardedPairEvaluator, shard_sizes, allgather_rows  # noqa: F401

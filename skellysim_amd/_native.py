@@ -46,6 +46,8 @@ def _bind(lib):
                                                           ctypes.c_double, ctypes.c_double]
     lib.skelly_stresslet_normal_density_device.argtypes = [pv, pv, pv, pv, _LL, _LL,
                                                             ctypes.c_double, ctypes.c_double, pv]
+    lib.skelly_stresslet_times_normal_device.argtypes = [pv, pv, pv, _LL, ctypes.c_double,
+                                                          ctypes.c_double, pv]
     lib.skelly_oseen_tensor_batched_device.argtypes = [pv, pv, _LL, _LL, ctypes.c_double,
                                                         ctypes.c_double, ctypes.c_double, pv]
     lib.skelly_fp64_peak_tflops.argtypes = [ctypes.POINTER(ctypes.c_double)]

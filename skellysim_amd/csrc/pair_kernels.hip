@@ -512,6 +512,62 @@ __global__ __launch_bounds__(BLOCK) void oseen_tensor_kernel(const double *__res
             blk[a * ld + b] = (a == b ? fr : 0.0) + gr * d[a] * d[b];
 }
 
+/* stresslet_times_normal dense builder (kernels.cpp:264-287):
+ * Snormal block (i, j) = -3/(4*pi) * (d.n_j)/r^5 * d d^T, d = r_i - r_j;
+ * i == j blocks zero (kernels.cpp:273-274); r < eps regularized
+ * (kernels.cpp:278-279). Output is (3n, 3n) C row-major with
+ * out[3i+a][3j+b] = Snormal(3i+a, 3j+b) — same logical indexing as the
+ * reference (its Eigen buffer is col-major; the MATRIX is identical). */
+__global__ __launch_bounds__(BLOCK) void stresslet_normal_kernel(
+    const double *__restrict__ pts, const double *__restrict__ normals, double *__restrict__ G,
+    long long n, double factor, double reg2, double eps2) {
+    const long long idx = (long long)blockIdx.x * BLOCK + threadIdx.x;
+    if (idx >= n * n)
+        return;
+    const long long i = idx / n, j = idx % n;
+    const long long ld = 3 * n;
+    double *blk = G + (3 * i) * ld + 3 * j;
+    if (i == j) {
+#pragma unroll
+        for (int a = 0; a < 3; ++a)
+#pragma unroll
+            for (int b = 0; b < 3; ++b)
+                blk[a * ld + b] = 0.0;
+        return;
+    }
+    const double dx = pts[3 * i + 0] - pts[3 * j + 0];
+    const double dy = pts[3 * i + 1] - pts[3 * j + 1];
+    const double dz = pts[3 * i + 2] - pts[3 * j + 2];
+    double dr2 = dx * dx;
+    dr2 = __builtin_fma(dy, dy, dr2);
+    dr2 = __builtin_fma(dz, dz, dr2);
+    const double denom2 = (dr2 < eps2) ? dr2 + reg2 : dr2;
+    const double y = rsq_refined(denom2);
+    const double y2 = y * y;
+    const double rinv5 = y * y2 * y2;
+    double ddn = dx * normals[3 * j + 0];
+    ddn = __builtin_fma(dy, normals[3 * j + 1], ddn);
+    ddn = __builtin_fma(dz, normals[3 * j + 2], ddn);
+    const double c = factor * ddn * rinv5;
+    const double d[3] = {dx, dy, dz};
+#pragma unroll
+    for (int a = 0; a < 3; ++a)
+#pragma unroll
+        for (int b = 0; b < 3; ++b)
+            blk[a * ld + b] = c * d[a] * d[b];
+}
+
+hipError_t launch_stresslet_times_normal(const double *pts, const double *normals, double *G,
+                                         long long n, double reg, double eps,
+                                         hipStream_t stream) {
+    if (n <= 0)
+        return hipSuccess;
+    const long long blocks = (n * n + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(stresslet_normal_kernel, dim3((unsigned)blocks), dim3(BLOCK), 0, stream,
+                       pts, normals, G, n, -3.0 / (4.0 * M_PI), reg * reg, eps * eps);
+    return hipGetLastError();
+}
+
 hipError_t launch_oseen_tensor_batched(const double *pts, double *G, long long nf, long long n,
                                        double eta, double reg, double eps, hipStream_t stream) {
     if (nf <= 0 || n <= 0)

@@ -32,6 +32,8 @@ hipError_t launch_stresslet_normal_density(const double *, const double *, const
                                            hipStream_t);
 hipError_t launch_oseen_tensor_batched(const double *, double *, long long, long long, double,
                                        double, double, hipStream_t);
+hipError_t launch_stresslet_times_normal(const double *, const double *, double *, long long,
+                                         double, double, hipStream_t);
 hipError_t run_fp64_peak(double *);
 } // namespace skelly
 
@@ -321,6 +323,15 @@ int skelly_stresslet_normal_density_device(const double *d_r_src, const double *
     CHK("skelly_stresslet_normal_density_device",
         skelly::launch_stresslet_normal_density(d_r_src, d_nd, d_r_trg, d_out, n_src, n_trg, reg,
                                                 epsilon_distance, (hipStream_t)stream));
+    return 0;
+}
+
+int skelly_stresslet_times_normal_device(const double *d_pts, const double *d_normals,
+                                         double *d_out, long long n, double reg,
+                                         double epsilon_distance, void *stream) {
+    CHK("skelly_stresslet_times_normal_device",
+        skelly::launch_stresslet_times_normal(d_pts, d_normals, d_out, n, reg,
+                                              epsilon_distance, (hipStream_t)stream));
     return 0;
 }
 

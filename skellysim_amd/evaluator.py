@@ -231,6 +231,22 @@ def stresslet_normal_density_device(r_src, normals, density, r_trg=None, reg=5e-
     return out
 
 
+def stresslet_times_normal_device(pts, normals, reg=5e-3, epsilon_distance=1e-5, out=None):
+    """Dense stresslet_times_normal builder (kernels.cpp:264-287).
+    pts, normals: (n, 3) CUDA fp64 -> (3n, 3n)."""
+    import torch
+    pts = _dev_rows(pts, 3, "pts")
+    normals = _dev_rows(normals, 3, "normals")
+    n = len(pts)
+    if out is None:
+        out = torch.empty((3 * n, 3 * n), dtype=torch.float64, device=pts.device)
+    rc = _native.lib().skelly_stresslet_times_normal_device(
+        ctypes.c_void_p(pts.data_ptr()), ctypes.c_void_p(normals.data_ptr()),
+        ctypes.c_void_p(out.data_ptr()), n, float(reg), float(epsilon_distance), _stream_ptr())
+    _native.check(rc, "stresslet_times_normal_device")
+    return out
+
+
 def oseen_tensor_batched_device(pts, eta=1.0, reg=5e-3, epsilon_distance=1e-5, out=None):
     """Batched dense self-Oseen-tensor builder (kernels.cpp:146-195; the
     per-fiber self-stokeslet, fiber_finite_difference.cpp:56).

@@ -168,6 +168,28 @@ def test_stresslet_normal_density_vs_oracle(ska, oracle_mod):
     assert np.array_equal(out3.cpu().numpy(), out)
 
 
+def test_stresslet_times_normal_dense_vs_oracle(ska, oracle_mod):
+    """Dense stresslet_times_normal builder (kernels.cpp:264-287) vs the
+    numpy restatement, plus the contraction identity
+    Snormal @ rho == stresslet_times_normal_times_density."""
+    import torch
+    rng = np.random.default_rng(61)
+    n = 300
+    r = rng.uniform(-1, 1, (n, 3))
+    nrm = rng.uniform(-1, 1, (n, 3))
+    rho = rng.uniform(-1, 1, (n, 3))
+    dev = torch.device("cuda:0")
+    S = ska.stresslet_times_normal_device(torch.from_numpy(r).to(dev),
+                                          torch.from_numpy(nrm).to(dev))
+    torch.cuda.synchronize()
+    Sh = S.cpu().numpy()
+    ref = oracle_mod.np_stresslet_times_normal(r, nrm)
+    assert rel(Sh, ref) < REL_TOL
+    contr = (Sh @ rho.reshape(-1)).reshape(-1, 3)
+    sdn = oracle_mod.stresslet_times_normal_times_density(r, nrm, rho)
+    assert rel(contr, sdn) < 1e-12
+
+
 def test_oseen_tensor_batched_vs_oracle(ska, oracle_mod):
     """Batched per-fiber self-stokeslet dense build (kernels.cpp:146-195 via
     fiber_finite_difference.cpp:56), incl. a near-regularized fiber."""
