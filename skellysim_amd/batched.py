@@ -12,6 +12,26 @@ factors resident in HBM across the solve.
 
 import torch
 
+_magma_latched = False
+
+
+def _lu_factor(mats):
+    """torch.linalg.lu_factor with a magma fallback: ROCm 7.2's
+    hipblasDgetrfBatched fails with HIPBLAS_STATUS_ALLOC_FAILED for m > 128
+    (measured on MI355X, profiles/components_r01.md); magma handles all the
+    fiber sizes (4n up to 256) at per-timestep-negligible cost."""
+    global _magma_latched
+    if not mats.is_cuda:
+        return torch.linalg.lu_factor(mats)
+    try:
+        return torch.linalg.lu_factor(mats)
+    except RuntimeError as e:
+        if "HIPBLAS" not in str(e) or _magma_latched:
+            raise
+        torch.backends.cuda.preferred_linalg_library("magma")
+        _magma_latched = True
+        return torch.linalg.lu_factor(mats)
+
 
 class BatchedLU:
     """Factor once per timestep, solve per GMRES iteration.
@@ -25,7 +45,7 @@ class BatchedLU:
             raise ValueError(f"expected (batch, m, m), got {tuple(mats.shape)}")
         if mats.dtype != torch.float64:
             raise TypeError("BatchedLU expects fp64")
-        self.LU, self.pivots = torch.linalg.lu_factor(mats)
+        self.LU, self.pivots = _lu_factor(mats)
 
     def solve(self, rhs):
         """rhs: (n_fibers, m) or (n_fibers, m, k) -> same shape solution."""
