@@ -121,9 +121,8 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
             y = torch.linalg.solve_triangular(H[:k_done, :k_done],
                                               g[:k_done].reshape(-1, 1),
                                               upper=True).reshape(-1)
-            update = torch.zeros_like(x)
-            for j in range(k_done):
-                update += y[j].item() * V[j]
+            basis = torch.stack(V[:k_done], dim=1)  # (n_local, k_done)
+            update = basis @ y.to(basis.device)
             x = x + precond(update)
         else:
             break
