@@ -41,7 +41,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=5)
     p.add_argument("--warmup", type=int, default=2)
-    p.add_argument("--n", type=int, default=1_000_000,
+    p.add_argument("--n", "--npoints", dest="n", type=int, default=1_000_000,
                    help="cloud size (sources == targets); default the metric point 1e6")
     p.add_argument("--seed", type=int, default=100)
     p.add_argument("--skip-cpu-baseline", action="store_true")
@@ -100,7 +100,11 @@ def main():
     args = parse_args()
     import torch
 
-    if not torch.cuda.is_available():
+    # CI rehearsal of the distributed plumbing (tests/test_bench_cli.py):
+    # gloo + CPU tensors + stub compute; never set by the driver.
+    rehearsal = os.environ.get("SKELLY_BENCH_REHEARSAL") == "gloo"
+
+    if not torch.cuda.is_available() and not rehearsal:
         print(json.dumps({"error": "no GPU visible; bench.py must run on an MI355X box"}))
         sys.exit(1)
 
@@ -119,8 +123,9 @@ def main():
         import torch.distributed as dist_mod
         dist = dist_mod
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        dist.init_process_group("nccl", rank=rank, world_size=world)
-    torch.cuda.set_device(local_rank)
+        dist.init_process_group("gloo" if rehearsal else "nccl", rank=rank, world_size=world)
+    if not rehearsal:
+        torch.cuda.set_device(local_rank)
 
     import skellysim_amd as ska
     from skellysim_amd import _native
@@ -135,17 +140,34 @@ def main():
 
     ts, te = shard_range(n, world, rank)   # target shard (contiguous blocks)
     ss, se = shard_range(n, world, rank)   # source shard (same split)
-    dev = torch.device("cuda", local_rank)
+    dev = torch.device("cpu") if rehearsal else torch.device("cuda", local_rank)
     r_src_local = torch.from_numpy(pts[ss:se].copy()).to(dev)
     f_src_local = torch.from_numpy(strengths[ss:se].copy()).to(dev)
     r_trg_local = torch.from_numpy(pts[ts:te].copy()).to(dev)
     u_local = torch.empty_like(r_trg_local)
-    torch.cuda.synchronize()
 
-    ev_start = [torch.cuda.Event(enable_timing=True) for _ in range(args.steps)]
-    ev_end = [torch.cuda.Event(enable_timing=True) for _ in range(args.steps)]
+    def sync():
+        if not rehearsal:
+            torch.cuda.synchronize()
 
-    if args.kernel == "stokeslet":
+    sync()
+
+    if rehearsal:
+        class _Ev:
+            def record(self):
+                pass
+            def elapsed_time(self, other):
+                return 1.0
+        ev_start = [_Ev() for _ in range(args.steps)]
+        ev_end = [_Ev() for _ in range(args.steps)]
+    else:
+        ev_start = [torch.cuda.Event(enable_timing=True) for _ in range(args.steps)]
+        ev_end = [torch.cuda.Event(enable_timing=True) for _ in range(args.steps)]
+
+    if rehearsal:
+        kernel_fn = lambda r, f, t, out: out.zero_()
+        flops_per_pair, metric = FLOPS_PER_PAIR, "rehearsal (no compute)"
+    elif args.kernel == "stokeslet":
         kernel_fn = lambda r, f, t, out: ska.stokeslet_device(r, f, t, eta, out=out)
         flops_per_pair, metric = FLOPS_PER_PAIR, "Stokeslet pair-interactions/sec (fp64)"
     else:  # regularized Stokeslet (BASELINE config 3); defaults kernels.hpp:34-35
@@ -166,15 +188,15 @@ def main():
 
     for _ in range(args.warmup):
         step()
-    torch.cuda.synchronize()
+    sync()
     if dist:
         dist.barrier()
-    torch.cuda.synchronize()
+    sync()
 
     t0 = time.perf_counter()
     for i in range(args.steps):
         step(i)
-    torch.cuda.synchronize()
+    sync()
     t1 = time.perf_counter()
     elapsed = t1 - t0
     if dist:
@@ -198,13 +220,14 @@ def main():
         traffic = load_traffic_calibration(n, te - ts) if args.kernel == "stokeslet" else None
 
         peak_meas = ctypes.c_double(0.0)
-        try:
-            _native.lib().skelly_fp64_peak_tflops(ctypes.byref(peak_meas))
-        except Exception:
-            pass
+        if not rehearsal:
+            try:
+                _native.lib().skelly_fp64_peak_tflops(ctypes.byref(peak_meas))
+            except Exception:
+                pass
 
         cpu = None
-        if world == 1 and not args.skip_cpu_baseline:
+        if world == 1 and not args.skip_cpu_baseline and not rehearsal:
             cpu = cpu_baseline(pts, strengths, pts, eta, kernel=args.kernel)
 
         out = {
