@@ -31,6 +31,7 @@ def assemble_shell_operator(nodes, normals, weights, eta=1.0, want_inverse=True)
     Returns (A, M_inv) device tensors (M_inv None if want_inverse=False)."""
     N = nodes.shape[0]
     dev = nodes.device
+    weights = weights.reshape(-1)  # the RBF tooling emits (N, 1)
 
     S = stresslet_times_normal_device(nodes, normals)  # (3N, 3N)
 
