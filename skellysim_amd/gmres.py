@@ -10,31 +10,30 @@ src/core/params.cpp:14). Operators are callables (the reference's
 A_fiber_hydro::apply / P_inv_hydro::apply just forward to System::
 apply_matvec / apply_preconditioner, solver_hydro.cpp:23-29,42-48).
 
-Distributed: vectors may be rank-local slices of a block-row-distributed
-global vector (the reference's Tpetra map, solver_hydro.cpp:17-20); inner
-products then all-reduce across ranks (Tpetra's distributed dots). The
-matvec/precond callables own whatever gathers they need (e.g.
-ShellOperator's source all-gather).
+The Krylov basis is one (n, restart+1) matrix and each ICGS pass is two
+GEMVs (h = V^T w; w -= V h) — one fused dot-block per pass instead of O(k)
+scalar ops (scalar torch CPU ops pay a fork-join on many-core hosts).
 
-Small dense Hessenberg algebra (size ~ iteration count) runs on the host —
-it is O(restart^2) scalars per iteration.
+Distributed: vectors may be rank-local slices of a block-row-distributed
+global vector (the reference's Tpetra map, solver_hydro.cpp:17-20); the
+dot-block and norms then all-reduce across ranks (Tpetra's distributed
+dots). The matvec/precond callables own whatever gathers they need.
 """
 
 import torch
 
 
-def _make_dot(distributed, group=None):
+def _make_reduce(distributed, group=None):
     if not distributed:
-        return lambda a, b: torch.dot(a, b)
+        return lambda t: t
 
     import torch.distributed as dist
 
-    def dot(a, b):
-        s = torch.dot(a, b)
-        dist.all_reduce(s, group=group)
-        return s
+    def allreduce(t):
+        dist.all_reduce(t, group=group)
+        return t
 
-    return dot
+    return allreduce
 
 
 def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
@@ -48,10 +47,13 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
     """
     if precond is None:
         precond = lambda v: v
-    dot = _make_dot(distributed, group)
-    norm = lambda v: torch.sqrt(dot(v, v))
+    reduce_ = _make_reduce(distributed, group)
+
+    def norm(v):
+        return torch.sqrt(reduce_(torch.dot(v, v)))
 
     b = b.reshape(-1)
+    n = b.shape[0]
     x = torch.zeros_like(b) if x0 is None else x0.clone().reshape(-1)
     bnorm = norm(b)
     if float(bnorm) == 0.0:
@@ -70,7 +72,8 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
             break
 
         m = min(restart, maxiter - total_iters)
-        V = [r / beta]
+        V = torch.zeros((n, m + 1), dtype=b.dtype, device=b.device)
+        V[:, 0] = r / beta
         H = torch.zeros((m + 1, m), dtype=torch.float64)
         g = torch.zeros(m + 1, dtype=torch.float64)
         g[0] = float(beta)
@@ -79,14 +82,16 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
         k_done = 0
 
         for k in range(m):
-            w = matvec(precond(V[k]))
+            w = matvec(precond(V[:, k].contiguous()))
             # ICGS: two classical Gram-Schmidt passes (Belos "ICGS",
-            # solver_hydro.cpp:72)
+            # solver_hydro.cpp:72), each as one fused dot-block + update
+            Vk = V[:, : k + 1]
+            hcol = torch.zeros(k + 1, dtype=torch.float64)
             for _ in range(2):
-                for j in range(k + 1):
-                    hjk = dot(V[j], w)
-                    H[j, k] += float(hjk)
-                    w = w - hjk * V[j]
+                h = reduce_(Vk.T @ w)
+                hcol += h.to(hcol.dtype).cpu()
+                w = w - Vk @ h
+            H[: k + 1, k] = hcol
             hk1 = norm(w)
             H[k + 1, k] = float(hk1)
 
@@ -115,14 +120,13 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
             if resid <= tol or float(hk1) == 0.0:
                 converged = resid <= tol
                 break
-            V.append(w / hk1)
+            V[:, k + 1] = w / hk1
 
         if k_done > 0:
             y = torch.linalg.solve_triangular(H[:k_done, :k_done],
                                               g[:k_done].reshape(-1, 1),
                                               upper=True).reshape(-1)
-            basis = torch.stack(V[:k_done], dim=1)  # (n_local, k_done)
-            update = basis @ y.to(basis.device)
+            update = V[:, :k_done] @ y.to(dtype=b.dtype, device=b.device)
             x = x + precond(update)
         else:
             break
