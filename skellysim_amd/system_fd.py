@@ -305,18 +305,129 @@ class SystemFD:
             res[self.fiber_sol_size:] = self._shell_precond(x[self.fiber_sol_size:])
         return res
 
+    # ---- device-resident iteration (uniform fibers + HipBackend) --------
+    def _build_device_operators(self):
+        """Stack the per-fiber operators in HBM so the entire GMRES iteration
+        (matvec + preconditioner) runs on device with zero host traffic —
+        SURVEY.md §8f row 3 in full."""
+        import torch
+        t = self.backend.torch
+        dev = self.backend.dev
+        T = lambda a: torch.from_numpy(np.ascontiguousarray(a)).to(dev)
+
+        nf = len(self.fibers)
+        f0 = self.fibers[0]
+        n = f0.n_nodes
+        self._dev = dict(nf=nf, n=n)
+        d = self._dev
+        d["A"] = T(np.stack([f.A for f in self.fibers]))
+        d["F"] = T(np.stack([f.force_operator for f in self.fibers]))
+        d["G"] = T(np.stack([f.stokeslet for f in self.fibers]))
+        d["xs"] = T(np.stack([f.xs for f in self.fibers]))          # (nf, 3, n)
+        d["w"] = T(np.concatenate([f.quadrature_weights() for f in self.fibers]))
+        d["P_ds"] = T(f0.mats["P_downsample_bc"])                    # (4n-14, 4n)
+        d["D1preT"] = T((f0.mats["D_1_0"] * (2.0 / f0.length_prev)).T)
+        d["r_fib"] = T(self.fiber_nodes())
+        d["r_all"] = T(self.all_nodes())
+        d["plus_vel"] = T(np.array(
+            [1.0 if f.bc_plus[0] == "Velocity" else 0.0 for f in self.fibers]))
+        if self.shell:
+            d["sh_nodes"] = T(self.shell.nodes)
+            d["sh_normals"] = T(self.shell.normals)
+            d["sh_A"] = self.shell.A if torch.is_tensor(self.shell.A) \
+                else T(self.shell.A)
+            d["sh_Minv"] = self.shell.M_inv if torch.is_tensor(self.shell.M_inv) \
+                else T(self.shell.M_inv)
+        from .batched import BatchedLU
+        d["lu"] = BatchedLU(d["A"])
+
+    def _apply_matvec_device(self, x):
+        """apply_matvec entirely on device (torch fp64 CUDA vector in/out)."""
+        import torch
+        from .evaluator import stokeslet_device, stresslet_device
+        d = self._dev
+        nf, n = d["nf"], d["n"]
+        eta = self.eta
+        nf_nodes = nf * n
+        x_fib = x[: 4 * nf_nodes].reshape(nf, 4 * n)
+        x_shell = x[4 * nf_nodes:]
+
+        # forces: F @ x (component-major) -> node-major (nf*n, 3)
+        fw = torch.bmm(d["F"], x_fib.unsqueeze(-1)).squeeze(-1)     # (nf, 3n)
+        fw_nodes = fw.reshape(nf, 3, n).permute(0, 2, 1).reshape(nf_nodes, 3)
+        wf = (fw_nodes * d["w"][:, None]).contiguous()
+
+        v_all = stokeslet_device(d["r_fib"], wf, d["r_all"], eta)
+        # per-fiber self subtraction (point-major, f_c_fd.cpp:203-210)
+        wf_pm = wf.reshape(nf, 3 * n, 1)
+        corr = torch.bmm(d["G"], wf_pm).reshape(nf_nodes, 3)
+        v_all[:nf_nodes] -= corr
+
+        if self.shell:
+            dens = x_shell.reshape(-1, 3)
+            f_dl = 2.0 * eta * torch.einsum("ni,nj->nij", d["sh_normals"],
+                                            dens).reshape(-1, 9).contiguous()
+            v_all[:nf_nodes] += stresslet_device(d["sh_nodes"], f_dl,
+                                                 d["r_fib"], eta)
+
+        # fiber matvec: A x - vT_in + xs_vT (fiber_fd.matvec, cpp:278-315)
+        v_fib = v_all[:nf_nodes].reshape(nf, n, 3).permute(0, 2, 1)  # (nf, 3, n)
+        vT = torch.zeros((nf, 4 * n), dtype=x.dtype, device=x.device)
+        vT[:, : 3 * n] = v_fib.reshape(nf, 3 * n)
+        tens = torch.einsum("ts,fis->fit", d["D1preT"],
+                            d["xs"] * v_fib).sum(dim=1)
+        vT[:, 3 * n:] = tens
+        vT_in = torch.zeros_like(vT)
+        vT_in[:, : 4 * n - 14] = vT @ d["P_ds"].T
+        res_fib = torch.bmm(d["A"], x_fib.unsqueeze(-1)).squeeze(-1) - vT_in
+        bc_start = 4 * n - 14
+        res_fib[:, bc_start + 3] += (v_fib[:, :, 0] * d["xs"][:, :, 0]).sum(dim=1)
+        res_fib[:, bc_start + 10] += d["plus_vel"] * \
+            (v_fib[:, :, -1] * d["xs"][:, :, -1]).sum(dim=1)
+
+        res = torch.empty_like(x)
+        res[: 4 * nf_nodes] = res_fib.reshape(-1)
+        if self.shell:
+            v_shell = v_all[nf_nodes:].reshape(-1)
+            res[4 * nf_nodes:] = d["sh_A"] @ x_shell + v_shell
+        return res
+
+    def _apply_precond_device(self, x):
+        import torch
+        d = self._dev
+        nf, n = d["nf"], d["n"]
+        res = torch.empty_like(x)
+        res[: 4 * nf * n] = d["lu"].solve(x[: 4 * nf * n].reshape(nf, 4 * n)).reshape(-1)
+        if self.shell:
+            res[4 * nf * n:] = d["sh_Minv"] @ x[4 * nf * n:]
+        return res
+
     def solve(self, tol=1e-10, maxiter=200, restart=None):
         """system.cpp:464-478 via the engine GMRES (right-preconditioned,
-        ICGS — solver_hydro.cpp:64-87)."""
+        ICGS — solver_hydro.cpp:64-87). With uniform fibers on the HIP
+        backend the whole iteration runs device-resident."""
         import torch
         from .gmres import gmres
 
         rhs = self.prep_state_for_solver()
+        if restart is None:
+            restart = min(200, maxiter)
+
+        device_mode = (self._uniform and self.fibers
+                       and isinstance(self.backend, HipBackend))
+        if device_mode:
+            self._build_device_operators()
+            b = self.backend._t(rhs)
+            x, info = gmres(self._apply_matvec_device, b,
+                            precond=self._apply_precond_device,
+                            tol=tol, maxiter=maxiter, restart=restart)
+            self.backend.torch.cuda.synchronize()
+            self.solution = x.cpu().numpy()
+            return info
+
         b = torch.from_numpy(rhs)
         mv = lambda v: torch.from_numpy(self.apply_matvec(v.numpy()))
         pc = lambda v: torch.from_numpy(self.apply_preconditioner(v.numpy()))
-        if restart is None:
-            restart = min(200, maxiter)
         x, info = gmres(mv, b, precond=pc, tol=tol, maxiter=maxiter, restart=restart)
         self.solution = x.numpy()
         return info
