@@ -30,6 +30,8 @@ class HipBackend:
         self.dev = torch.device(device)
 
     def _t(self, a):
+        if self.torch.is_tensor(a):
+            return a.to(self.dev)
         return self.torch.from_numpy(np.ascontiguousarray(a)).to(self.dev)
 
     def stokeslet(self, r_src, f_src, r_trg, eta):
