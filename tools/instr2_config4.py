@@ -39,3 +39,12 @@ t0=time.perf_counter()
 _, info = gmres(nullmv, b, tol=0, maxiter=50, restart=50)
 torch.cuda.synchronize()
 print(f"gmres-internal per iter (null op, device): {(time.perf_counter()-t0)/max(1,info['iters'])*1e3:.2f} ms  iters={info['iters']}")
+
+import cProfile, pstats, io
+pr = cProfile.Profile()
+pr.enable()
+info = sys_.solve(tol=1e-10, maxiter=40, restart=40)
+pr.disable()
+s = io.StringIO(); pstats.Stats(pr, stream=s).sort_stats("tottime").print_stats(16)
+print("iters:", info["iters"])
+print(s.getvalue())
