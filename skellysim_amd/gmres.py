@@ -20,6 +20,7 @@ dot-block and norms then all-reduce across ranks (Tpetra's distributed
 dots). The matvec/precond callables own whatever gathers they need.
 """
 
+import numpy as np
 import torch
 
 
@@ -74,11 +75,13 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
         m = min(restart, maxiter - total_iters)
         V = torch.zeros((n, m + 1), dtype=b.dtype, device=b.device)
         V[:, 0] = r / beta
-        H = torch.zeros((m + 1, m), dtype=torch.float64)
-        g = torch.zeros(m + 1, dtype=torch.float64)
+        # small dense Hessenberg/Givens state in numpy: scalar torch-CPU ops
+        # cost a fork-join on many-core hosts
+        H = np.zeros((m + 1, m))
+        g = np.zeros(m + 1)
         g[0] = float(beta)
-        cs = torch.zeros(m, dtype=torch.float64)
-        sn = torch.zeros(m, dtype=torch.float64)
+        cs = np.zeros(m)
+        sn = np.zeros(m)
         k_done = 0
 
         for k in range(m):
@@ -86,21 +89,23 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
             # ICGS: two classical Gram-Schmidt passes (Belos "ICGS",
             # solver_hydro.cpp:72), each as one fused dot-block + update
             Vk = V[:, : k + 1]
-            hcol = torch.zeros(k + 1, dtype=torch.float64)
+            hcol = np.zeros(k + 1)
             for _ in range(2):
                 h = reduce_(Vk.T @ w)
-                hcol += h.to(hcol.dtype).cpu()
                 w = w - Vk @ h
+                hcol += h.cpu().numpy()
             H[: k + 1, k] = hcol
             hk1 = norm(w)
             H[k + 1, k] = float(hk1)
 
-            # Givens rotations on the new column
-            for j in range(k):
-                t = cs[j] * H[j, k] + sn[j] * H[j + 1, k]
-                H[j + 1, k] = -sn[j] * H[j, k] + cs[j] * H[j + 1, k]
-                H[j, k] = t
-            denom = torch.sqrt(H[k, k] ** 2 + H[k + 1, k] ** 2)
+            # Givens rotations on the new column (vectorized in numpy)
+            if k:
+                col = H[: k + 1, k]
+                for j in range(k):
+                    t = cs[j] * col[j] + sn[j] * col[j + 1]
+                    col[j + 1] = -sn[j] * col[j] + cs[j] * col[j + 1]
+                    col[j] = t
+            denom = np.sqrt(H[k, k] ** 2 + H[k + 1, k] ** 2)
             if float(denom) == 0.0:
                 k_done = k
                 break
@@ -123,10 +128,10 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
             V[:, k + 1] = w / hk1
 
         if k_done > 0:
-            y = torch.linalg.solve_triangular(H[:k_done, :k_done],
-                                              g[:k_done].reshape(-1, 1),
-                                              upper=True).reshape(-1)
-            update = V[:, :k_done] @ y.to(dtype=b.dtype, device=b.device)
+            import scipy.linalg as _scla
+            y = _scla.solve_triangular(H[:k_done, :k_done], g[:k_done])
+            yt = torch.from_numpy(y).to(dtype=b.dtype, device=b.device)
+            update = V[:, :k_done] @ yt
             x = x + precond(update)
         else:
             break

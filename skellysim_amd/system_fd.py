@@ -182,6 +182,16 @@ class SystemFD:
         per-fiber self term on the fiber's own slice of the targets."""
         if not self.fibers:
             return np.zeros_like(r_trg)
+        if self.fibers[0].stokeslet is None:
+            if self._uniform:
+                pts = np.stack([f.x.T for f in self.fibers])
+                G = self.backend.self_stokeslet_batch(pts, self.eta)
+                for f, g in zip(self.fibers, G):
+                    f.stokeslet = g
+            else:
+                for f in self.fibers:
+                    f.stokeslet = self.backend.self_stokeslet_batch(
+                        f.x.T[None], self.eta)[0]
         w = np.concatenate([f.quadrature_weights() for f in self.fibers])
         wf = fib_forces * w[:, None]
         vel = self.backend.stokeslet(self.fiber_nodes(), wf, r_trg, self.eta)
@@ -214,17 +224,11 @@ class SystemFD:
             f.update_linear_operator(dt, eta)
             f.update_force_operator()
 
-        # self-stokeslets (fiber_finite_difference.cpp:56), batched on device
-        if self.fibers:
-            if self._uniform:
-                pts = np.stack([f.x.T for f in self.fibers])
-                G = self.backend.self_stokeslet_batch(pts, eta)
-                for f, g in zip(self.fibers, G):
-                    f.stokeslet = g
-            else:
-                for f in self.fibers:
-                    f.stokeslet = self.backend.self_stokeslet_batch(
-                        f.x.T[None], eta)[0]
+        # self-stokeslets (fiber_finite_difference.cpp:56) are built lazily:
+        # the device-resident solve path builds its own resident stack, and
+        # the host path materializes them on first _fiber_flow use
+        for f in self.fibers:
+            f.stokeslet = None
 
         r_all = self.all_nodes()
         nf_nodes = self.fiber_node_count
@@ -247,9 +251,10 @@ class SystemFD:
             f.apply_bc_rectangular(dt, v_fib[a:b].T, None)
 
         # preconditioner: batched LU of the (BC-applied) fiber operators
+        # (lazy — the device-resident solve path factors its own resident
+        # copy in _build_device_operators)
         if self.fibers and self._uniform:
-            A_batch = np.stack([f.A for f in self.fibers])
-            self._fiber_lu_solve = self.backend.batched_lu(A_batch)
+            self._fiber_lu_solve = None
         elif self.fibers:
             solves = [self.backend.batched_lu(f.A[None]) for f in self.fibers]
             self._fiber_lu_solve = None
@@ -296,6 +301,9 @@ class SystemFD:
         x_fib = x[: self.fiber_sol_size]
         if self.fibers:
             if self._uniform:
+                if self._fiber_lu_solve is None:
+                    A_batch = np.stack([f.A for f in self.fibers])
+                    self._fiber_lu_solve = self.backend.batched_lu(A_batch)
                 m = 4 * self.fibers[0].n_nodes
                 sol = self._fiber_lu_solve(x_fib.reshape(len(self.fibers), m))
                 res[: self.fiber_sol_size] = sol.reshape(-1)
@@ -324,7 +332,10 @@ class SystemFD:
         d = self._dev
         d["A"] = T(np.stack([f.A for f in self.fibers]))
         d["F"] = T(np.stack([f.force_operator for f in self.fibers]))
-        d["G"] = T(np.stack([f.stokeslet for f in self.fibers]))
+        # self-stokeslets built directly on device (kernels.cpp:146-195)
+        from .evaluator import oseen_tensor_batched_device
+        pts = T(np.stack([f.x.T for f in self.fibers]))
+        d["G"] = oseen_tensor_batched_device(pts, eta=self.eta)
         d["xs"] = T(np.stack([f.xs for f in self.fibers]))          # (nf, 3, n)
         d["w"] = T(np.concatenate([f.quadrature_weights() for f in self.fibers]))
         d["P_ds"] = T(f0.mats["P_downsample_bc"])                    # (4n-14, 4n)

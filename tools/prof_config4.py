@@ -34,7 +34,7 @@ def main(n_fibers=512, n_nodes=64):
     A, M_inv = assemble_shell_operator(nodes_t, normals_t, w_t)
     torch.cuda.synchronize()
     t_shell = time.perf_counter() - t0
-    shell = Shell(fix["nodes"], fix["normals"], A.cpu().numpy(), M_inv.cpu().numpy())
+    shell = Shell(fix["nodes"], fix["normals"], A, M_inv)  # device-resident
     print(f"shell operator (8192 nodes): {t_shell:.2f}s (on-GPU assembly + inversion)")
 
     # fibers: length 1.0, rigidity 2.5e-3 (survey config-4 params), minus end
