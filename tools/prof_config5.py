@@ -1,0 +1,98 @@
+#!/usr/bin/env python3
+"""BASELINE config-5 shape on one MI355X: ~4k fibers x 32 nodes,
+minus-clamped inside a 6000-node spherical periphery (oocyte-scale), N
+backward-Euler timesteps with the device-resident GMRES, plus a
+velocity-field cross-check of the final state against the CPU oracle
+(config 5's "velocity-field tolerance vs CPU").
+
+The 8-GPU leg of config 5 is round-2 scope; this measures the single-GPU
+timestep throughput of the full pipeline."""
+
+import argparse
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np
+import torch
+
+from skellysim_amd.fiber_fd import FiberFD
+from skellysim_amd.system_fd import SystemFD, HipBackend, Shell
+from skellysim_amd.periphery_precompute import assemble_shell_operator
+from skellysim_amd.flows import velocity_at_targets
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--fibers", type=int, default=4000)
+    ap.add_argument("--nodes", type=int, default=32)
+    ap.add_argument("--steps", type=int, default=5)
+    args = ap.parse_args()
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    fix = np.load(os.path.join(repo, "tests", "golden", "sphere_6000_nodes.npz"))
+    dev = torch.device("cuda:0")
+
+    t0 = time.perf_counter()
+    A, M_inv = assemble_shell_operator(torch.from_numpy(fix["nodes"]).to(dev),
+                                       torch.from_numpy(fix["normals"]).to(dev),
+                                       torch.from_numpy(fix["quadrature_weights"]).to(dev))
+    torch.cuda.synchronize()
+    print(f"shell operator (6000 nodes): {time.perf_counter()-t0:.2f}s")
+    shell = Shell(fix["nodes"], fix["normals"], A, M_inv)
+
+    length, E = 1.0, 2.5e-3
+    sel = np.linspace(0, len(fix["nodes"]) - 1, args.fibers).astype(int)
+    fibers = []
+    for i in sel:
+        p = fix["nodes"][i]
+        n = fix["normals"][i] / np.linalg.norm(fix["normals"][i])
+        s = np.linspace(0.05, 0.05 + length, args.nodes)
+        fibers.append(FiberFD(p[None, :] + s[:, None] * n[None, :], length=length,
+                              bending_rigidity=E, eta=1.0, minus_clamped=True,
+                              force_scale=-0.05))
+
+    sys_ = SystemFD(fibers, eta=1.0, dt=0.025, shell=shell, backend=HipBackend())
+    print(f"solution size: {sys_.fiber_sol_size + sys_.shell_sol_size}")
+
+    t0 = time.perf_counter()
+    iters = []
+    for k in range(args.steps):
+        info = sys_.step(tol=1e-10, maxiter=300, restart=150)
+        iters.append(info["iters"])
+        assert info["converged"], info
+    dt_total = time.perf_counter() - t0
+    print(f"{args.steps} timesteps: {dt_total:.2f}s = {dt_total/args.steps:.2f} s/step "
+          f"({args.steps/dt_total:.3f} steps/s); GMRES iters per step: {iters}")
+
+    # velocity-field cross-check vs the CPU oracle on the FINAL state
+    import oracle
+    rng = np.random.default_rng(2)
+    pts = rng.uniform(-0.5, 0.5, (64, 3)) * float(fix["radius"])
+    r_fib = sys_.fiber_nodes()
+    w = np.concatenate([f.quadrature_weights() for f in fibers])
+    fw = np.zeros_like(r_fib)
+    off = 0
+    for f in fibers:  # motor forces as the field source (force_scale * xs)
+        fw[off: off + f.n_nodes] = (f.force_scale * f.xs).T
+        off += f.n_nodes
+    dens = sys_.solution[sys_.fiber_sol_size:].reshape(-1, 3)
+
+    T = lambda a: torch.from_numpy(np.ascontiguousarray(a)).to(dev)
+    u_gpu = velocity_at_targets(
+        T(pts), 1.0,
+        fiber=dict(r_src=T(r_fib), forces=T(fw), weights=T(w)),
+        shell=dict(node_pos=T(fix["nodes"]), node_normal=T(fix["normals"]),
+                   density=T(dens)))
+    torch.cuda.synchronize()
+    f_dl = 2.0 * np.einsum("ni,nj->nij", fix["normals"], dens).reshape(-1, 9)
+    u_cpu = (oracle.stokeslet(r_fib, fw * w[:, None], pts, 1.0)
+             + oracle.stresslet(fix["nodes"], f_dl, pts, 1.0))
+    rel = np.linalg.norm(u_gpu.cpu().numpy() - u_cpu) / np.linalg.norm(u_cpu)
+    print(f"velocity field at 64 interior targets: GPU vs CPU oracle rel = {rel:.3e}")
+
+
+if __name__ == "__main__":
+    main()
