@@ -33,6 +33,20 @@ def _lu_factor(mats):
         return torch.linalg.lu_factor(mats)
 
 
+def robust_inv(A):
+    """torch.linalg.inv with the same magma fallback (hipblasDtrsm also hits
+    HIPBLAS_STATUS_ALLOC_FAILED at some sizes on ROCm 7.2)."""
+    global _magma_latched
+    try:
+        return torch.linalg.inv(A)
+    except RuntimeError as e:
+        if "HIPBLAS" not in str(e) or _magma_latched:
+            raise
+        torch.backends.cuda.preferred_linalg_library("magma")
+        _magma_latched = True
+        return torch.linalg.inv(A)
+
+
 class BatchedLU:
     """Factor once per timestep, solve per GMRES iteration.
 

@@ -59,5 +59,6 @@ def assemble_shell_operator(nodes, normals, weights, eta=1.0, want_inverse=True)
     nflat = normals.reshape(-1)
     A = S + torch.outer(nflat, nflat)
 
-    M_inv = torch.linalg.inv(A) if want_inverse else None
+    from .batched import robust_inv
+    M_inv = robust_inv(A) if want_inverse else None
     return A, M_inv
