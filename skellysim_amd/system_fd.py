@@ -445,9 +445,9 @@ class SystemFD:
         self.solution = x.numpy()
         return info
 
-    def step(self, tol=1e-10, maxiter=200):
+    def step(self, tol=1e-10, maxiter=200, restart=None):
         """system.cpp:482-493 (no bodies): solve then adopt positions."""
-        info = self.solve(tol=tol, maxiter=maxiter)
+        info = self.solve(tol=tol, maxiter=maxiter, restart=restart)
         for f, a, b in self._fiber_slices():
             f.step(self.solution[a:b])
         return info
