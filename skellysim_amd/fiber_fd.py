@@ -381,6 +381,11 @@ class FiberFD:
         self.A[4 * np_ - 14:, :] = B
         self.RHS[4 * np_ - 14:] = B_RHS
 
+    def adopt_operator(self, A, RHS):
+        """Install externally (batch-)assembled operator state."""
+        self.A = A
+        self.RHS = RHS
+
     def quadrature_weights(self):
         """0.5 * length * weights_0 (fiber_container_finite_difference.cpp:186)."""
         return 0.5 * self.length * self.mats["weights_0"]

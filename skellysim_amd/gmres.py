@@ -87,16 +87,24 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
         for k in range(m):
             w = matvec(precond(V[:, k].contiguous()))
             # ICGS: two classical Gram-Schmidt passes (Belos "ICGS",
-            # solver_hydro.cpp:72), each as one fused dot-block + update
+            # solver_hydro.cpp:72), each as one fused dot-block + update.
+            # All scalars stay on device until ONE combined host transfer
+            # per iteration (each .cpu()/float() is a full-stream sync).
             Vk = V[:, : k + 1]
-            hcol = np.zeros(k + 1)
+            hcol_dev = None
             for _ in range(2):
                 h = reduce_(Vk.T @ w)
                 w = w - Vk @ h
-                hcol += h.cpu().numpy()
-            H[: k + 1, k] = hcol
-            hk1 = norm(w)
-            H[k + 1, k] = float(hk1)
+                hcol_dev = h if hcol_dev is None else hcol_dev + h
+            hk1_dev = torch.sqrt(reduce_(torch.dot(w, w)))
+            # device-side normalize without a host read; if hk1 == 0 the
+            # column is never consumed (the loop breaks below)
+            V[:, k + 1] = w / hk1_dev
+            host_scalars = torch.cat([hcol_dev.reshape(-1),
+                                      hk1_dev.reshape(1)]).cpu().numpy()
+            H[: k + 1, k] = host_scalars[:-1]
+            hk1 = host_scalars[-1]
+            H[k + 1, k] = hk1
 
             # Givens rotations on the new column (vectorized in numpy)
             if k:
@@ -106,7 +114,7 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
                     col[j + 1] = -sn[j] * col[j] + cs[j] * col[j + 1]
                     col[j] = t
             denom = np.sqrt(H[k, k] ** 2 + H[k + 1, k] ** 2)
-            if float(denom) == 0.0:
+            if denom == 0.0:
                 k_done = k
                 break
             cs[k] = H[k, k] / denom
@@ -122,10 +130,9 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
             residuals.append(resid)
             if callback is not None:
                 callback(total_iters, resid)
-            if resid <= tol or float(hk1) == 0.0:
+            if resid <= tol or hk1 == 0.0:
                 converged = resid <= tol
                 break
-            V[:, k + 1] = w / hk1
 
         if k_done > 0:
             import scipy.linalg as _scla

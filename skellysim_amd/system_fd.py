@@ -221,8 +221,10 @@ class SystemFD:
         for f in self.fibers:
             f.update_constants(eta)
             f.update_derivatives()
-            f.update_linear_operator(dt, eta)
-            f.update_force_operator()
+        if not (self._uniform and self.fibers):
+            for f in self.fibers:
+                f.update_linear_operator(dt, eta)
+                f.update_force_operator()
 
         # self-stokeslets (fiber_finite_difference.cpp:56) are built lazily:
         # the device-resident solve path builds its own resident stack, and
@@ -246,9 +248,19 @@ class SystemFD:
             v_all += self.background_flow(r_all)
 
         v_fib = v_all[:nf_nodes]
-        for f, a, b in self._fiber_node_slices():
-            f.update_RHS(dt, v_fib[a:b].T, motor[a:b].T)
-            f.apply_bc_rectangular(dt, v_fib[a:b].T, None)
+        if self._uniform and self.fibers:
+            # batched assembly (operator + RHS + BCs + force operator) —
+            # numerically identical to the per-fiber loop (fiber_batch.py)
+            from .fiber_batch import assemble_uniform
+            n = self.fibers[0].n_nodes
+            nf = len(self.fibers)
+            flow_b = v_fib.reshape(nf, n, 3).transpose(0, 2, 1)
+            motor_b = motor.reshape(nf, n, 3).transpose(0, 2, 1)
+            assemble_uniform(self.fibers, dt, eta, flow=flow_b, f_external=motor_b)
+        else:
+            for f, a, b in self._fiber_node_slices():
+                f.update_RHS(dt, v_fib[a:b].T, motor[a:b].T)
+                f.apply_bc_rectangular(dt, v_fib[a:b].T, None)
 
         # preconditioner: batched LU of the (BC-applied) fiber operators
         # (lazy — the device-resident solve path factors its own resident
