@@ -1,0 +1,104 @@
+"""Reference-format trajectory output: writes the
+msgpack stream SkellySim's own Python tooling (src/skelly_sim/reader.py)
+reads, so the reference's analysis/paraview ecosystem consumes this engine's
+output unchanged.
+
+Format contract (verified against the reference sources):
+  header  — header_map_t (include/io_maps.hpp:44-57): map {trajversion=1
+            (CMakeLists.txt:12), number_mpi_ranks, fiber_type,
+            skellysim_version, skellysim_commit, simdate, hostname}
+  frame   — output_map_t (io_maps.hpp:31-41): map {time, dt, rng_state,
+            fibers, bodies, shell}
+    fibers — FiberContainerFiniteDifference, packed as the ARRAY
+             [fiber_type_, [fiber maps]] (MSGPACK_DEFINE,
+             fiber_container_finite_difference.hpp:126; FIBERTYPE
+             FiniteDifference == 1, fiber_container_base.hpp:19)
+    fiber  — map with the keys of fiber_finite_difference.hpp:160-161
+    bodies — [spherical, deformable, ellipsoidal] lists (body_container.hpp:158)
+    shell  — map {solution_vec_} (periphery.hpp:120)
+    Eigen matrices — ['__eigen__', rows, cols, column-major values]
+             (include/eigen_matrix_plugin.h:30-41)
+"""
+
+import datetime
+import socket
+
+import msgpack
+import numpy as np
+
+TRAJECTORY_VERSION = 1  # reference CMakeLists.txt:12
+FIBERTYPE_FINITEDIFFERENCE = 1  # fiber_container_base.hpp:19
+
+
+def eigen(a):
+    """['__eigen__', rows, cols, col-major...] (eigen_matrix_plugin.h).
+    (n, 3) numpy arrays become 3 x n Eigen matrices (point-major values are
+    identical); 1-D arrays become n x 1 vectors."""
+    a = np.asarray(a, dtype=np.float64)
+    if a.ndim == 1:
+        return ["__eigen__", a.size, 1] + a.tolist()
+    if a.ndim == 2 and a.shape[1] == 3:
+        return ["__eigen__", 3, a.shape[0]] + a.reshape(-1).tolist()
+    if a.ndim == 2:  # general: emit as (rows, cols) col-major
+        return ["__eigen__", a.shape[0], a.shape[1]] + a.T.reshape(-1).tolist()
+    raise ValueError(f"unsupported shape {a.shape}")
+
+
+def fiber_map(f):
+    """fiber_finite_difference.hpp:160-161 (non-resume field set)."""
+    return {
+        "n_nodes_": f.n_nodes,
+        "radius_": f.radius,
+        "length_": f.length,
+        "length_prev_": f.length_prev,
+        "bending_rigidity_": f.bending_rigidity,
+        "penalty_param_": f.penalty_param,
+        "force_scale_": f.force_scale,
+        "beta_tstep_": f.beta_tstep,
+        "binding_site_": [-1, -1],
+        "tension_": eigen(f.tension),
+        "x_": eigen(f.x.T),  # (n, 3) -> 3 x n Eigen
+        "minus_clamped_": f.minus_clamped,
+    }
+
+
+class TrajectoryWriter:
+    """Streams reference-format frames for a SystemFD run."""
+
+    def __init__(self, path, version="skelly-hip-0.1.0", commit="unknown"):
+        self._fh = open(path, "wb")
+        header = {
+            "trajversion": TRAJECTORY_VERSION,
+            "number_mpi_ranks": 1,
+            "fiber_type": FIBERTYPE_FINITEDIFFERENCE,
+            "skellysim_version": version,
+            "skellysim_commit": commit,
+            "simdate": datetime.datetime.now().isoformat(),
+            "hostname": socket.gethostname(),
+        }
+        self._fh.write(msgpack.packb(header))
+
+    def write_frame(self, system, time, dt):
+        shell_sol = (system.solution[system.fiber_sol_size:]
+                     if getattr(system, "solution", None) is not None
+                     and system.shell else np.zeros(0))
+        frame = {
+            "time": float(time),
+            "dt": float(dt),
+            "rng_state": [],
+            "fibers": [FIBERTYPE_FINITEDIFFERENCE,
+                       [fiber_map(f) for f in system.fibers]],
+            "bodies": [[], [], []],
+            "shell": {"solution_vec_": eigen(np.asarray(shell_sol))},
+        }
+        self._fh.write(msgpack.packb(frame))
+        self._fh.flush()
+
+    def close(self):
+        self._fh.close()
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        self.close()

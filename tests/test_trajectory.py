@@ -1,0 +1,118 @@
+"""Trajectory output in the reference's msgpack format (§8f row 4):
+structural checks against the serialization contract, and — when the
+reference tree is present (build container) — a round trip through the
+REFERENCE'S OWN TrajectoryReader (src/skelly_sim/reader.py)."""
+
+import os
+import sys
+import types
+
+import msgpack
+import numpy as np
+import pytest
+
+from skellysim_amd.fiber_fd import FiberFD
+from skellysim_amd.system_fd import SystemFD, OracleBackend
+from skellysim_amd.trajectory import TrajectoryWriter
+
+REFERENCE = "/root/reference"
+
+
+def _small_system():
+    rng = np.random.default_rng(4)
+    s = np.linspace(0, 1.0, 16)
+    d = np.array([0.0, 0.0, 1.0])
+    fibers = [FiberFD(np.array([0.1 * k, 0, 0])[None, :] + s[:, None] * d[None, :],
+                      length=1.0, bending_rigidity=2.5e-3, eta=1.0)
+              for k in range(2)]
+    U = np.array([0.1, 0.0, 0.0])
+    return SystemFD(fibers, eta=1.0, dt=0.1, backend=OracleBackend(),
+                    background_flow=lambda r: np.tile(U, (len(r), 1)))
+
+
+def _write_traj(dirpath, steps=2):
+    sys_ = _small_system()
+    path = os.path.join(dirpath, "skelly_sim.out")
+    times = []
+    with TrajectoryWriter(path) as tw:
+        t = 0.0
+        for _ in range(steps):
+            info = sys_.step(tol=1e-11, maxiter=200)
+            assert info["converged"]
+            t += sys_.dt
+            times.append(t)
+            tw.write_frame(sys_, t, sys_.dt)
+    return path, times, sys_
+
+
+def test_structure_matches_serialization_contract(tmp_path):
+    path, times, sys_ = _write_traj(str(tmp_path))
+    with open(path, "rb") as fh:
+        unp = msgpack.Unpacker(fh, raw=False)
+        header = next(unp)
+        assert header["trajversion"] == 1
+        assert header["fiber_type"] == 1
+        frames = list(unp)
+    assert len(frames) == len(times)
+    fr = frames[-1]
+    assert set(fr.keys()) == {"time", "dt", "rng_state", "fibers", "bodies", "shell"}
+    assert fr["time"] == pytest.approx(times[-1])
+    ftype, fibs = fr["fibers"]
+    assert ftype == 1 and len(fibs) == 2
+    f0 = fibs[0]
+    assert f0["n_nodes_"] == 16
+    x = f0["x_"]
+    assert x[0] == "__eigen__" and x[1] == 3 and x[2] == 16
+    got = np.array(x[3:]).reshape(16, 3)
+    assert np.allclose(got, sys_.fibers[0].x.T)
+    assert fr["bodies"] == [[], [], []]
+
+
+@pytest.mark.skipif(not os.path.isdir(REFERENCE), reason="reference tree absent")
+def test_reference_reader_round_trip(tmp_path):
+    """The REFERENCE'S OWN reader.py must load our trajectory: times indexed,
+    frames decoded, fiber positions identical."""
+    # shims for the reference python package's optional deps
+    for name, mod in {
+        "toml": dict(load=lambda f: {}, dump=lambda *a, **k: None),
+        "dataclass_utils": dict(check_type=lambda *a, **k: None),
+        "numba": dict(njit=lambda *a, **k: (a[0] if a and callable(a[0])
+                                            else (lambda f: f)), prange=range),
+    }.items():
+        m = types.ModuleType(name)
+        for k, v in mod.items():
+            setattr(m, k, v)
+        sys.modules.setdefault(name, m)
+    if "nptyping" not in sys.modules:
+        m = types.ModuleType("nptyping")
+
+        class _Sub:
+            def __class_getitem__(cls, item):
+                return np.ndarray
+
+        m.NDArray = _Sub
+        m.Shape = _Sub
+        m.Float64 = float
+        sys.modules["nptyping"] = m
+    fg = types.ModuleType("function_generator")
+    fg.FunctionGenerator = type("FunctionGenerator", (), {"__init__": lambda s, *a, **k: None})
+    sys.modules.setdefault("function_generator", fg)
+
+    sys.path.insert(0, os.path.join(REFERENCE, "src"))
+    try:
+        from skelly_sim.reader import TrajectoryReader
+    except Exception as e:
+        pytest.skip(f"reference reader not importable here: {e}")
+
+    path, times, sys_ = _write_traj(str(tmp_path))
+    toml_file = os.path.join(str(tmp_path), "skelly_config.toml")
+    open(toml_file, "w").write("")
+
+    traj = TrajectoryReader(toml_file)
+    assert traj.trajectory_version == 1
+    assert list(traj.times) == pytest.approx(times)
+    traj.load_frame(len(times) - 1)
+    fibs = traj["fibers"]
+    assert len(fibs) == 2
+    assert np.allclose(fibs[0]["x_"], sys_.fibers[0].x.T)
+    assert traj["time"] == pytest.approx(times[-1])
