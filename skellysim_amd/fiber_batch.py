@@ -9,7 +9,7 @@ Validated block-for-block against the per-fiber path (tests/test_fiber_fd.py).
 
 import numpy as np
 
-from .fiber_fd import BC_VELOCITY, BC_ANGULAR_VELOCITY, BC_FORCE, BC_TORQUE
+from .fiber_fd import BC_VELOCITY
 
 
 def assemble_uniform(fibers, dt, eta, flow=None, f_external=None):

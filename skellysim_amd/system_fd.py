@@ -18,8 +18,6 @@ validate the orchestration end-to-end off-GPU).
 
 import numpy as np
 
-from .fiber_fd import FiberFD
-
 
 class HipBackend:
     """Product backend: pair kernels + batched algebra on the MI355X."""

@@ -75,7 +75,6 @@ def test_hip_matches_oracle_backend_one_solve(hip_backend):
     """One multi-fiber + small-shell solve: HIP backend equals the oracle
     backend to the GMRES tolerance."""
     import os
-    import torch
     from skellysim_amd.system_fd import SystemFD, OracleBackend, Shell
     from skellysim_amd.fiber_fd import FiberFD
 
