@@ -113,8 +113,7 @@ def assemble_uniform(fibers, dt, eta, flow=None, f_external=None):
     RHS[:, : 4 * n - 14] = RHS @ P.T
     B = np.zeros((nf, 14, 4 * n))
     B_RHS = np.zeros((nf, 14))
-    von = flow  # v_on_fiber == background+induced flow slice (prep passes it)
-    fon = None  # external (periphery) forces are zero in this harness
+    von = flow  # v_on_fiber (prep passes it); external forces f_on_fiber are zero here
 
     # group fibers by BC signature (vectorize within each group)
     minus_vel = np.array([f.bc_minus[0] == BC_VELOCITY for f in fibers])
