@@ -1,0 +1,180 @@
+"""Listener service — the reference's post-processing protocol
+(src/core/listener.cpp) served by this engine: size-prefixed msgpack
+requests on stdin, responses on stdout, against a trajectory file.
+
+Wire contract (listener.cpp:86-137 / reader.py Listener.request):
+  request  — uint64-LE size + msgpack map {frame_no, evaluator,
+             streamlines{...}, vortexlines{...}, velocity_field{x}}
+             (arrays __eigen__-encoded by the reference client)
+  response — uint64-LE size + msgpack map {time, i_frame, n_frames,
+             streamlines, vortexlines, velocity_field}
+  size 0 in → terminate; invalid frame -> size-0 response (listener.cpp:110-115)
+
+Velocity-field semantics mirror System::velocity_at_targets
+(system.cpp:330-384) for fibers + shell: fiber forces are
+force_operator @ fiber_solution (apply_fiber_force on the frame's
+positions+tension), fiber flow WITHOUT self-subtraction (flow(..., false),
+system.cpp:355), plus the shell double layer from the frame's
+solution_vec_. Streamlines/vortexlines are not implemented (empty lists;
+the reference integrates them with Boost odeint — out of scope)."""
+
+import argparse
+import struct
+import sys
+
+import msgpack
+import numpy as np
+
+from .fiber_fd import FiberFD
+
+
+def eigen_decode(d):
+    """reader.py::_eigen_to_numpy (same rules)."""
+    if isinstance(d, list):
+        if d and d[0] == "__eigen__":
+            if d[1] == 1 or d[2] == 1:
+                return np.array(d[3:])
+            if d[1] == 3:
+                return np.array(d[3:]).reshape((d[2], d[1]))
+            return np.array(d[3:]).reshape((d[2], d[1])).transpose()
+        return [eigen_decode(x) for x in d]
+    if isinstance(d, dict):
+        return {k: eigen_decode(v) for k, v in d.items()}
+    return d
+
+
+def eigen_encode_3xn(a):
+    """(n, 3) -> ['__eigen__', 3, n, colmajor] (eigen_matrix_plugin.h)."""
+    a = np.asarray(a, dtype=np.float64)
+    return ["__eigen__", 3, len(a), *a.reshape(-1).tolist()]
+
+
+class Trajectory:
+    """Minimal reader of the reference-format trajectory stream."""
+
+    def __init__(self, path):
+        self.frames = []
+        with open(path, "rb") as fh:
+            unp = msgpack.Unpacker(fh, raw=False)
+            first = next(unp)
+            if isinstance(first, dict) and "trajversion" in first:
+                self.header = first
+            else:
+                self.header = {}
+                self.frames.append(eigen_decode(first))
+            for obj in unp:
+                self.frames.append(eigen_decode(obj))
+
+    def __len__(self):
+        return len(self.frames)
+
+
+def fibers_from_frame(frame, eta):
+    ftype, fmaps = frame["fibers"]
+    fibers = []
+    for fm in fmaps:
+        f = FiberFD(fm["x_"], length=fm["length_"],
+                    bending_rigidity=fm["bending_rigidity_"], eta=eta,
+                    radius=fm["radius_"], force_scale=fm["force_scale_"],
+                    minus_clamped=bool(fm["minus_clamped_"]),
+                    penalty_param=fm["penalty_param_"],
+                    beta_tstep=fm["beta_tstep_"])
+        f.length_prev = fm["length_prev_"]
+        t = np.asarray(fm["tension_"], float).reshape(-1)
+        f.tension = t if t.size == f.n_nodes else np.zeros(f.n_nodes)
+        f.update_derivatives()
+        f.update_force_operator()
+        fibers.append(f)
+    return fibers
+
+
+def velocity_field(frame, targets, eta, compute, shell_geometry=None):
+    """System::velocity_at_targets for fibers (+shell when geometry given).
+    compute: object with stokeslet/stresslet_normal_density (a system_fd
+    backend)."""
+    u = np.zeros((len(targets), 3))
+    fibers = fibers_from_frame(frame, eta)
+    if fibers:
+        r_src, wf = [], []
+        for f in fibers:
+            sol = np.concatenate([f.x.reshape(-1), f.tension])
+            ff = f.force_operator @ sol  # apply_fiber_force, system.cpp:339
+            fn = np.stack([ff[i * f.n_nodes:(i + 1) * f.n_nodes]
+                           for i in range(3)], axis=1)
+            r_src.append(f.x.T)
+            wf.append(fn * f.quadrature_weights()[:, None])
+        # fc_->flow(..., subtract_self=false), system.cpp:355
+        u += compute.stokeslet(np.concatenate(r_src), np.concatenate(wf),
+                               targets, eta)
+    if shell_geometry is not None and "shell" in frame:
+        dens = np.asarray(frame["shell"].get("solution_vec_", np.zeros(0)),
+                          float).reshape(-1)
+        if dens.size:
+            u += compute.stresslet_normal_density(
+                shell_geometry["nodes"], shell_geometry["normals"],
+                dens.reshape(-1, 3), targets, eta)
+    return u
+
+
+def serve(stdin, stdout, traj, compute, eta=1.0, shell_geometry=None):
+    """The stdin/stdout request loop (listener.cpp:86-137)."""
+    while True:
+        raw = stdin.read(8)
+        if len(raw) < 8:
+            return
+        (msgsize,) = struct.unpack("<Q", raw)
+        if msgsize == 0:
+            return
+        payload = b""
+        while len(payload) < msgsize:
+            chunk = stdin.read(msgsize - len(payload))
+            if not chunk:
+                return
+            payload += chunk
+        cmd = msgpack.unpackb(payload, raw=False)
+        frame_no = int(cmd.get("frame_no", 0))
+        if frame_no < 0 or frame_no >= len(traj):
+            stdout.write(struct.pack("<Q", 0))
+            stdout.flush()
+            continue
+        frame = traj.frames[frame_no]
+        vf = cmd.get("velocity_field", {}) or {}
+        x = eigen_decode(vf.get("x", []))
+        x = np.asarray(x, float).reshape(-1, 3) if np.size(x) else np.zeros((0, 3))
+        u = velocity_field(frame, x, eta, compute, shell_geometry) \
+            if len(x) else np.zeros((0, 3))
+        response = {
+            "time": float(frame["time"]),
+            "i_frame": frame_no,
+            "n_frames": len(traj),
+            "streamlines": [],
+            "vortexlines": [],
+            "velocity_field": eigen_encode_3xn(u),
+        }
+        out = msgpack.packb(response)
+        stdout.write(struct.pack("<Q", len(out)))
+        stdout.write(out)
+        stdout.flush()
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--trajectory", default="skelly_sim.out")
+    ap.add_argument("--eta", type=float, default=1.0)
+    ap.add_argument("--shell-geometry", default=None,
+                    help="npz with nodes/normals (the precompute geometry)")
+    args = ap.parse_args()
+
+    from .system_fd import HipBackend  # product path: the MI355X engine
+    compute = HipBackend()
+    shell_geometry = None
+    if args.shell_geometry:
+        fx = np.load(args.shell_geometry)
+        shell_geometry = {"nodes": fx["nodes"], "normals": fx["normals"]}
+    traj = Trajectory(args.trajectory)
+    serve(sys.stdin.buffer, sys.stdout.buffer, traj, compute, eta=args.eta,
+          shell_geometry=shell_geometry)
+
+
+if __name__ == "__main__":
+    main()

@@ -78,38 +78,6 @@ class HipBackend:
         return matvec, precond
 
 
-class OracleBackend:
-    """TEST-ONLY backend over the CPU oracle (oracle/ is test infrastructure;
-    this class exists so the orchestration logic is CPU-testable and must
-    never be used outside tests)."""
-
-    def __init__(self):
-        import oracle
-        self.oracle = oracle
-
-    def stokeslet(self, r_src, f_src, r_trg, eta):
-        return self.oracle.stokeslet(r_src, f_src, r_trg, eta)
-
-    def stresslet_normal_density(self, r_src, normals, density, r_trg, eta):
-        f_dl = 2.0 * eta * np.einsum("ni,nj->nij", normals, density).reshape(-1, 9)
-        return self.oracle.stresslet(r_src, f_dl, r_trg, eta)
-
-    def self_stokeslet_batch(self, pts, eta):
-        return np.stack([self.oracle.oseen_tensor(p, eta) for p in pts])
-
-    def batched_lu(self, A_batch):
-        import scipy.linalg as scla
-        lus = [scla.lu_factor(A) for A in A_batch]
-
-        def solve(rhs):
-            return np.stack([scla.lu_solve(lu, r) for lu, r in zip(lus, rhs)])
-
-        return solve
-
-    def shell_ops(self, A, M_inv):
-        return (lambda x: A @ x), (lambda x: M_inv @ x)
-
-
 class Shell:
     """Periphery state: nodes/normals (N, 3) + the dense operators
     (stresslet_plus_complementary, M_inv)."""

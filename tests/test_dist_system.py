@@ -36,7 +36,8 @@ def _make_problem():
 def _dist_worker(rank, world, init_file, q):
     import torch.distributed as dist
     from skellysim_amd.system_dist import DistributedSystemFD, distribute_fibers
-    from skellysim_amd.system_fd import OracleBackend, Shell
+    from skellysim_amd.system_fd import Shell
+    from oracle_backend import OracleBackend
     from skellysim_amd.sharded import shard_range
 
     dist.init_process_group("gloo", init_method=f"file://{init_file}",
@@ -78,7 +79,8 @@ def test_distributed_solve_matches_single_process():
             assert p.exitcode == 0
 
     # single-process reference solve
-    from skellysim_amd.system_fd import SystemFD, OracleBackend, Shell
+    from skellysim_amd.system_fd import SystemFD, Shell
+    from oracle_backend import OracleBackend
     fx, fibers, bg = _make_problem()
     shell = Shell(fx["nodes"], fx["normals"], fx["stresslet_plus_complementary"],
                   fx["M_inv"])

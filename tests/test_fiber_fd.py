@@ -8,7 +8,8 @@ import numpy as np
 import pytest
 
 from skellysim_amd.fiber_fd import FiberFD, finite_diff, barycentric_matrix, fib_mats
-from skellysim_amd.system_fd import SystemFD, OracleBackend
+from skellysim_amd.system_fd import SystemFD
+from oracle_backend import OracleBackend
 
 
 def test_finite_diff_polynomial_exactness():

@@ -75,7 +75,8 @@ def test_hip_matches_oracle_backend_one_solve(hip_backend):
     """One multi-fiber + small-shell solve: HIP backend equals the oracle
     backend to the GMRES tolerance."""
     import os
-    from skellysim_amd.system_fd import SystemFD, OracleBackend, Shell
+    from skellysim_amd.system_fd import SystemFD, Shell
+    from oracle_backend import OracleBackend
     from skellysim_amd.fiber_fd import FiberFD
 
     here = os.path.dirname(os.path.abspath(__file__))

@@ -1,0 +1,111 @@
+"""CPU tests of the listener protocol (skellysim_amd/listener.py) against the
+reference's wire format: requests encoded exactly as reader.py's
+Listener.request does (__eigen__ ndencode), responses decoded with its
+_eigen_to_numpy rules."""
+
+import io
+import struct
+
+import msgpack
+import numpy as np
+import pytest
+
+from skellysim_amd.fiber_fd import FiberFD
+from skellysim_amd.system_fd import SystemFD
+from oracle_backend import OracleBackend
+from skellysim_amd.trajectory import TrajectoryWriter
+from skellysim_amd.listener import Trajectory, serve, eigen_decode
+
+
+def _ndencode(obj):  # reader.py:15-19
+    if isinstance(obj, np.ndarray):
+        return ["__eigen__", obj.shape[1], obj.shape[0]] + obj.ravel().tolist()
+    return obj
+
+
+def _request_bytes(frame_no, x):
+    cmd = {
+        "frame_no": frame_no,
+        "evaluator": "GPU",
+        "streamlines": {"dt_init": 0.1, "t_final": 1.0, "abs_err": 1e-10,
+                        "rel_err": 1e-6, "back_integrate": True,
+                        "x0": np.zeros((0, 3))},
+        "vortexlines": {"dt_init": 0.1, "t_final": 1.0, "abs_err": 1e-10,
+                        "rel_err": 1e-6, "back_integrate": True,
+                        "x0": np.zeros((0, 3))},
+        "velocity_field": {"x": np.asarray(x, float)},
+    }
+    msg = msgpack.packb(cmd, default=_ndencode)
+    return struct.pack("<Q", len(msg)) + msg
+
+
+def _write_traj(tmp_path, steps=2):
+    s = np.linspace(0, 1.0, 16)
+    fibers = [FiberFD(np.array([0.2 * k, 0, 0])[None, :] +
+                      s[:, None] * np.array([0.0, 0, 1.0])[None, :],
+                      length=1.0, bending_rigidity=2.5e-3, eta=1.0,
+                      force_scale=-0.02)
+              for k in range(2)]
+    sys_ = SystemFD(fibers, eta=1.0, dt=0.1, backend=OracleBackend(),
+                    background_flow=lambda r: np.tile([0.1, 0, 0], (len(r), 1)))
+    path = str(tmp_path / "skelly_sim.out")
+    with TrajectoryWriter(path) as tw:
+        t = 0.0
+        for _ in range(steps):
+            assert sys_.step(tol=1e-11, maxiter=200)["converged"]
+            t += sys_.dt
+            tw.write_frame(sys_, t, sys_.dt)
+    return path, sys_
+
+
+def _roundtrip(path, requests, compute):
+    traj = Trajectory(path)
+    stdin = io.BytesIO(b"".join(requests) + struct.pack("<Q", 0))
+    stdout = io.BytesIO()
+    serve(stdin, stdout, traj, compute, eta=1.0)
+    stdout.seek(0)
+    responses = []
+    while True:
+        raw = stdout.read(8)
+        if len(raw) < 8:
+            break
+        (size,) = struct.unpack("<Q", raw)
+        if size == 0:
+            responses.append(None)
+            continue
+        responses.append(msgpack.unpackb(stdout.read(size), raw=False))
+    return responses
+
+
+def test_velocity_field_request(tmp_path):
+    path, sys_ = _write_traj(tmp_path)
+    targets = np.array([[0.5, 0.3, 0.5], [1.0, -0.2, 0.1], [0.0, 0.0, 2.0]])
+    reqs = [_request_bytes(1, targets)]
+    (res,) = _roundtrip(path, reqs, OracleBackend())
+    assert res["i_frame"] == 1 and res["n_frames"] == 2
+    assert res["time"] == pytest.approx(0.2)
+    u = eigen_decode(res["velocity_field"])
+    assert u.shape == (3, 3)
+
+    # semantics: flow of force_operator@solution, quadrature-weighted, NO
+    # self subtraction (system.cpp:339,355) — recompute directly
+    import oracle
+    r_src, wf = [], []
+    for f in sys_.fibers:
+        sol = np.concatenate([f.x.reshape(-1), f.tension])
+        ff = f.force_operator @ sol
+        fn = np.stack([ff[i * f.n_nodes:(i + 1) * f.n_nodes] for i in range(3)],
+                      axis=1)
+        r_src.append(f.x.T)
+        wf.append(fn * f.quadrature_weights()[:, None])
+    ref = oracle.stokeslet(np.concatenate(r_src), np.concatenate(wf), targets, 1.0)
+    assert np.allclose(u, ref, atol=1e-12)
+
+
+def test_invalid_frame_gives_empty_response(tmp_path):
+    path, _ = _write_traj(tmp_path)
+    reqs = [_request_bytes(99, np.zeros((1, 3))), _request_bytes(0, np.zeros((1, 3)))]
+    res = _roundtrip(path, reqs, OracleBackend())
+    assert res[0] is None          # size-0 response (listener.cpp:110-115)
+    assert res[1] is not None      # listener continues serving
+    assert res[1]["i_frame"] == 0

@@ -12,7 +12,8 @@ import numpy as np
 import pytest
 
 from skellysim_amd.fiber_fd import FiberFD
-from skellysim_amd.system_fd import SystemFD, OracleBackend
+from skellysim_amd.system_fd import SystemFD
+from oracle_backend import OracleBackend
 from skellysim_amd.trajectory import TrajectoryWriter
 
 REFERENCE = "/root/reference"
