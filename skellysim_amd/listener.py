@@ -15,8 +15,9 @@ Velocity-field semantics mirror System::velocity_at_targets
 force_operator @ fiber_solution (apply_fiber_force on the frame's
 positions+tension), fiber flow WITHOUT self-subtraction (flow(..., false),
 system.cpp:355), plus the shell double layer from the frame's
-solution_vec_. Streamlines/vortexlines are not implemented (empty lists;
-the reference integrates them with Boost odeint — out of scope)."""
+solution_vec_. Streamlines are integrated per the reference's adaptive
+5(4) RK scheme (see integrate_streamline); vortex lines (curl-field lines)
+are not implemented (empty list)."""
 
 import argparse
 import struct
@@ -116,6 +117,67 @@ def velocity_field(frame, targets, eta, compute, shell_geometry=None):
     return u
 
 
+def integrate_streamline(field_fn, x0, dt_init=0.1, t_final=1.0, abs_err=1e-10,
+                         rel_err=1e-6, back_integrate=True):
+    """One streamline (StreamLine::compute, src/core/streamline.cpp:66-118):
+    adaptive 5(4) Runge-Kutta integration of dx/dt = u(x) forward to t_final
+    (and backward when requested), with the reference's |u| > 1e3 singularity
+    stop. The reference uses Boost odeint's Cash-Karp 5(4); scipy's RK45
+    (Dormand-Prince 5(4)) integrates the same ODE to the same tolerances —
+    step placement differs, the curve does not.
+
+    Returns dict {x: (npts, 3), val: (npts, 3), time: [..]} (streamline.hpp:29).
+    """
+    from scipy.integrate import solve_ivp
+
+    def rhs(t, x):
+        return field_fn(x.reshape(1, 3)).reshape(3)
+
+    def singularity(t, x):
+        return 1e3 - np.linalg.norm(rhs(t, x))
+
+    singularity.terminal = True
+
+    def run(t_end, first):
+        sol = solve_ivp(rhs, (0.0, t_end), np.asarray(x0, float).reshape(3),
+                        method="RK45", first_step=abs(first), atol=abs_err,
+                        rtol=rel_err, events=singularity, dense_output=False)
+        return sol.t, sol.y.T
+
+    t_fwd, x_fwd = run(t_final, dt_init)
+    if back_integrate:
+        t_back, x_back = run(-t_final, dt_init)
+        # join: reversed backward path (minus the duplicate seed) + forward
+        # (streamline.cpp:55-64)
+        t = np.concatenate([t_back[:0:-1], t_fwd])
+        x = np.concatenate([x_back[:0:-1], x_fwd])
+    else:
+        t, x = t_fwd, x_fwd
+    val = np.stack([field_fn(p.reshape(1, 3)).reshape(3) for p in x])
+    return {"x": x, "val": val, "time": t.tolist()}
+
+
+def process_streamlines(frame, req, eta, compute, shell_geometry):
+    req = req or {}
+    x0 = eigen_decode(req.get("x0", []))
+    x0 = np.asarray(x0, float).reshape(-1, 3) if np.size(x0) else np.zeros((0, 3))
+    if not len(x0):
+        return []
+    field = lambda pts: velocity_field(frame, pts, eta, compute, shell_geometry)
+    out = []
+    for seed in x0:
+        s = integrate_streamline(field, seed,
+                                 dt_init=float(req.get("dt_init", 0.1)),
+                                 t_final=float(req.get("t_final", 1.0)),
+                                 abs_err=float(req.get("abs_err", 1e-10)),
+                                 rel_err=float(req.get("rel_err", 1e-6)),
+                                 back_integrate=bool(req.get("back_integrate", True)))
+        out.append({"x": eigen_encode_3xn(s["x"]),
+                    "val": eigen_encode_3xn(s["val"]),
+                    "time": s["time"]})
+    return out
+
+
 def serve(stdin, stdout, traj, compute, eta=1.0, shell_geometry=None):
     """The stdin/stdout request loop (listener.cpp:86-137)."""
     while True:
@@ -147,8 +209,9 @@ def serve(stdin, stdout, traj, compute, eta=1.0, shell_geometry=None):
             "time": float(frame["time"]),
             "i_frame": frame_no,
             "n_frames": len(traj),
-            "streamlines": [],
-            "vortexlines": [],
+            "streamlines": process_streamlines(frame, cmd.get("streamlines"),
+                                               eta, compute, shell_geometry),
+            "vortexlines": [],  # curl-field lines: out of scope this round
             "velocity_field": eigen_encode_3xn(u),
         }
         out = msgpack.packb(response)

@@ -102,6 +102,51 @@ def test_velocity_field_request(tmp_path):
     assert np.allclose(u, ref, atol=1e-12)
 
 
+def test_streamline_request(tmp_path):
+    """Streamlines: seeds integrate along the frame's velocity field
+    (midpoint finite differences of the returned path match the returned
+    velocities), back+forward paths joined with monotone time. Uses a BENT
+    fiber (a straight unloaded fiber exerts zero bending force -> zero
+    field)."""
+    s = np.linspace(0, 1.0, 24)
+    x = np.stack([0.15 * np.sin(2 * np.pi * s), np.zeros_like(s), s], axis=1)
+    fib = FiberFD(x, length=1.0, bending_rigidity=2.5e-2, eta=1.0)
+    sys_ = SystemFD([fib], eta=1.0, dt=0.1, backend=OracleBackend())
+    path = str(tmp_path / "skelly_sim.out")
+    with TrajectoryWriter(path) as tw:
+        tw.write_frame(sys_, 0.1, 0.1)
+    cmd = {
+        "frame_no": 0,
+        "evaluator": "GPU",
+        "streamlines": {"dt_init": 0.05, "t_final": 0.5, "abs_err": 1e-10,
+                        "rel_err": 1e-8, "back_integrate": True,
+                        "x0": np.array([[0.6, 0.2, 0.4]])},
+        "vortexlines": {"dt_init": 0.1, "t_final": 1.0, "abs_err": 1e-10,
+                        "rel_err": 1e-6, "back_integrate": True,
+                        "x0": np.zeros((0, 3))},
+        "velocity_field": {"x": np.zeros((0, 3))},
+    }
+    msg = msgpack.packb(cmd, default=_ndencode)
+    (res,) = _roundtrip(path, [struct.pack("<Q", len(msg)) + msg], OracleBackend())
+    assert len(res["streamlines"]) == 1
+    sl = res["streamlines"][0]
+    x = eigen_decode(sl["x"])
+    val = eigen_decode(sl["val"])
+    t = np.asarray(sl["time"])
+    assert x.shape == val.shape and x.shape[1] == 3 and len(t) == len(x)
+    assert np.all(np.diff(t) > 0) and t[0] < 0 < t[-1]
+    # the seed lies on the path at t == 0
+    i0 = np.argmin(np.abs(t))
+    assert np.allclose(x[i0], [0.6, 0.2, 0.4], atol=1e-12)
+    # midpoint finite differences track the reported velocities
+    dxdt = (x[2:] - x[:-2]) / (t[2:] - t[:-2])[:, None]
+    mid = val[1:-1]
+    mask = np.linalg.norm(mid, axis=1) > 1e-8
+    relerr = np.linalg.norm(dxdt[mask] - mid[mask], axis=1) / \
+        np.linalg.norm(mid[mask], axis=1)
+    assert np.median(relerr) < 0.25
+
+
 def test_invalid_frame_gives_empty_response(tmp_path):
     path, _ = _write_traj(tmp_path)
     reqs = [_request_bytes(99, np.zeros((1, 3))), _request_bytes(0, np.zeros((1, 3)))]
