@@ -12,10 +12,12 @@ import numpy as np
 from .fiber_fd import BC_VELOCITY
 
 
-def assemble_uniform(fibers, dt, eta, flow=None, f_external=None):
+def assemble_uniform(fibers, dt, eta, flow=None, f_external=None, bc_force=None):
     """update_linear_operator + update_RHS + apply_bc_rectangular for a
-    uniform-fiber population. flow/f_external: (nf, 3, n) or None.
-    Installs A/RHS on each fiber (adopt_operator)."""
+    uniform-fiber population. flow/f_external/bc_force: (nf, 3, n) or None
+    (bc_force is the f_on_fiber the reference passes to apply_bcs — the
+    EXTERNAL forces only, system.cpp:453). Installs A/RHS on each fiber
+    (adopt_operator)."""
     nf = len(fibers)
     f0 = fibers[0]
     n = f0.n_nodes
@@ -113,7 +115,8 @@ def assemble_uniform(fibers, dt, eta, flow=None, f_external=None):
     RHS[:, : 4 * n - 14] = RHS @ P.T
     B = np.zeros((nf, 14, 4 * n))
     B_RHS = np.zeros((nf, 14))
-    von = flow  # v_on_fiber (prep passes it); external forces f_on_fiber are zero here
+    von = flow  # v_on_fiber (prep passes it)
+    fon = bc_force
 
     # group fibers by BC signature (vectorize within each group)
     minus_vel = np.array([f.bc_minus[0] == BC_VELOCITY for f in fibers])
@@ -133,6 +136,8 @@ def assemble_uniform(fibers, dt, eta, flow=None, f_external=None):
         B_RHS[idx, 0:3] = x[idx, :, 0] / dt
         if von is not None:
             B_RHS[idx, 3] -= (xs[idx, :, 0] * von[idx, :, 0]).sum(axis=1)
+        if fon is not None:
+            B_RHS[idx, 3] -= 2 * c0f[idx] * (xs[idx, :, 0] * fon[idx, :, 0]).sum(axis=1)
     idx = np.where(~minus_vel)[0]
     if len(idx):
         for i in range(3):
@@ -141,7 +146,9 @@ def assemble_uniform(fibers, dt, eta, flow=None, f_external=None):
             B[idx, 3, i * n: (i + 1) * n] = \
                 (-Ef[idx] * xss[idx, i, 0])[:, None] * D2[idx, 0]
         B[idx, 3, 3 * n] = -1.0
-        # f_on_fiber is zero -> B_RHS rows stay zero
+        if fon is not None:
+            B_RHS[idx, 0:3] = fon[idx, :, 0]
+            B_RHS[idx, 3] = (fon[idx, :, 0] * xs[idx, :, 0]).sum(axis=1)
     # minus second BC
     idx = np.where(minus_vel)[0]  # AngularVelocity pairs with Velocity here
     if len(idx):
@@ -165,6 +172,8 @@ def assemble_uniform(fibers, dt, eta, flow=None, f_external=None):
         B_RHS[idx, 7:10] = x[idx, :, -1] / dt
         if von is not None:
             B_RHS[idx, 10] -= (xs[idx, :, -1] * von[idx, :, -1]).sum(axis=1)
+        if fon is not None:
+            B_RHS[idx, 10] -= 2 * c0f[idx] * (xs[idx, :, -1] * fon[idx, :, -1]).sum(axis=1)
     idx = np.where(~plus_vel)[0]  # Force
     if len(idx):
         for i in range(3):
@@ -173,6 +182,9 @@ def assemble_uniform(fibers, dt, eta, flow=None, f_external=None):
             B[idx, 10, i * n: (i + 1) * n] = \
                 (Ef[idx] * xss[idx, i, -1])[:, None] * D2[idx, -1]
         B[idx, 10, 4 * n - 1] = 1.0
+        if fon is not None:
+            B_RHS[idx, 7:10] = fon[idx, :, -1]
+            B_RHS[idx, 10] = (fon[idx, :, -1] * xs[idx, :, -1]).sum(axis=1)
     # plus second BC: Torque for all supported configurations
     for i in range(3):
         B[:, 11 + i, i * n: (i + 1) * n] = D2[:, -1]

@@ -381,6 +381,48 @@ class FiberFD:
         self.A[4 * np_ - 14:, :] = B
         self.RHS[4 * np_ - 14:] = B_RHS
 
+    def periphery_repulsion(self, kind, f_0=20.0, l_0=0.05, radius=None,
+                            abc=None):
+        """Steric repulsion from the periphery on this fiber's nodes
+        (SphericalPeriphery::fiber_interaction, periphery.cpp:140-162;
+        EllipsoidalPeriphery::fiber_interaction, periphery.cpp:232-263;
+        defaults f_0=20, l_0=0.05, params.hpp:46-47). Returns (3, n);
+        minus-clamped fibers skip node 0 (periphery.cpp:148)."""
+        pc = self.x
+        f = np.zeros_like(pc)
+        start = 1 if self.minus_clamped else 0
+        for i in range(start, pc.shape[1]):
+            p = pc[:, i]
+            r_mag = np.linalg.norm(p)
+            if kind == "sphere":
+                if r_mag >= radius or r_mag == 0.0:
+                    # >= radius: collision, force routine leaves zero
+                    # (periphery.cpp:152); == 0: the reference would divide
+                    # 0/0 — a node exactly at the center gets no force here
+                    continue
+                u_hat = p / r_mag
+                dr = p - u_hat * radius
+                gap = radius - r_mag
+            elif kind == "ellipsoid":
+                a, b, c = abc
+                r_s = p / np.array([a, b, c])
+                r_s_mag = np.linalg.norm(r_s)
+                phi = np.arctan2(r_s[1], r_s[0] + 1e-12)
+                theta = np.arccos(r_s[2] / (1e-12 + r_s_mag))
+                st = np.sin(theta)
+                r_cortex = np.array([a * st * np.cos(phi), b * st * np.sin(phi),
+                                     c * np.cos(theta)])
+                r_cortex_mag = np.linalg.norm(r_cortex)
+                if r_mag >= r_cortex_mag:
+                    continue
+                dr = p - r_cortex
+                gap = r_cortex_mag - r_mag
+            else:
+                raise ValueError(kind)
+            d = np.linalg.norm(dr)
+            f[:, i] = f_0 * dr / d * np.exp(-gap / l_0)
+        return f
+
     def adopt_operator(self, A, RHS):
         """Install externally (batch-)assembled operator state."""
         self.A = A

@@ -91,12 +91,16 @@ class Shell:
 
 
 class SystemFD:
-    def __init__(self, fibers, eta, dt, shell=None, background_flow=None, backend=None):
+    def __init__(self, fibers, eta, dt, shell=None, background_flow=None, backend=None,
+                 periphery_interaction=None):
         self.fibers = list(fibers)
         self.eta = float(eta)
         self.dt = float(dt)
         self.shell = shell
         self.background_flow = background_flow  # fn: (n,3) -> (n,3)
+        # steric fiber-periphery repulsion (system.cpp:421, params.cpp:18):
+        # dict(kind="sphere"|"ellipsoid", f_0=, l_0=, radius=|abc=) or None
+        self.periphery_interaction = periphery_interaction
         self.backend = backend if backend is not None else HipBackend()
         self._uniform = all(f.n_nodes == self.fibers[0].n_nodes for f in self.fibers) \
             if self.fibers else True
@@ -206,13 +210,24 @@ class SystemFD:
         for f, a, b in self._fiber_node_slices():
             motor[a:b] = (f.force_scale * f.xs).T
 
-        # v_all: background flow only (no point/body sources here;
-        # external/periphery-interaction forces are zero, so fc_->flow
-        # contributes nothing — system.cpp:425)
+        # fiber-periphery steric repulsion (fc_->periphery_force,
+        # system.cpp:421; per-fiber force periphery.cpp:140-162,232-263)
+        ext = np.zeros((nf_nodes, 3))
+        if self.periphery_interaction is not None:
+            pi = self.periphery_interaction
+            for f, a, b in self._fiber_node_slices():
+                ext[a:b] = f.periphery_repulsion(**pi).T
+
+        # v_all: flow induced by the external (periphery) forces
+        # (fc_->flow(r_all, external_force_fibers), system.cpp:425)
+        # + background flow; motor forces enter only the RHS
         v_all = np.zeros_like(r_all)
+        if self.periphery_interaction is not None and nf_nodes:
+            v_all += self._fiber_flow(r_all, ext)
         if self.background_flow is not None:
             v_all += self.background_flow(r_all)
 
+        motor = motor + ext  # total_force_fibers (system.cpp:450)
         v_fib = v_all[:nf_nodes]
         if self._uniform and self.fibers:
             # batched assembly (operator + RHS + BCs + force operator) —
@@ -222,11 +237,16 @@ class SystemFD:
             nf = len(self.fibers)
             flow_b = v_fib.reshape(nf, n, 3).transpose(0, 2, 1)
             motor_b = motor.reshape(nf, n, 3).transpose(0, 2, 1)
-            assemble_uniform(self.fibers, dt, eta, flow=flow_b, f_external=motor_b)
+            ext_b = ext.reshape(nf, n, 3).transpose(0, 2, 1) \
+                if self.periphery_interaction is not None else None
+            assemble_uniform(self.fibers, dt, eta, flow=flow_b, f_external=motor_b,
+                             bc_force=ext_b)
         else:
             for f, a, b in self._fiber_node_slices():
                 f.update_RHS(dt, v_fib[a:b].T, motor[a:b].T)
-                f.apply_bc_rectangular(dt, v_fib[a:b].T, None)
+                f.apply_bc_rectangular(dt, v_fib[a:b].T, ext[a:b].T
+                                       if self.periphery_interaction is not None
+                                       else None)
 
         # preconditioner: batched LU of the (BC-applied) fiber operators
         # (lazy — the device-resident solve path factors its own resident

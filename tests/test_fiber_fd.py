@@ -100,6 +100,44 @@ def test_batched_assembly_matches_per_fiber_loop():
         assert np.allclose(RHS[k], Rr, atol=1e-13, rtol=1e-13), k
         assert np.allclose(F[k], Fr, atol=1e-13, rtol=1e-13), k
 
+    # with external bc forces (the f_on_fiber leg of apply_bc_rectangular)
+    bc_f = [rng.uniform(-1, 1, (3, n)) for _ in fibers]
+    refs2 = []
+    for f, fl, mo, bf in zip(fibers, flows, motors, bc_f):
+        f.update_linear_operator(dt, eta)
+        f.update_RHS(dt, fl, mo)
+        f.apply_bc_rectangular(dt, fl, bf)
+        refs2.append(f.RHS.copy())
+    _, RHS2, _ = assemble_uniform(fibers, dt, eta, flow=np.stack(flows),
+                                  f_external=np.stack(motors),
+                                  bc_force=np.stack(bc_f))
+    for k, Rr in enumerate(refs2):
+        assert np.allclose(RHS2[k], Rr, atol=1e-13, rtol=1e-13), k
+
+
+def test_periphery_repulsion_force():
+    """Steric wall force (periphery.cpp:140-162): inward-pointing, decays
+    with the gap, zero outside/collided, clamped node skipped."""
+    n = 8
+    s = np.linspace(0.0, 0.9, n)
+    fib = straight_fiber(n=n, length=0.9, direction=(1.0, 0, 0))
+    f = fib.periphery_repulsion("sphere", radius=1.0)
+    # forces point OUTWARD toward the wall... dr = p - u_hat*radius points
+    # inward (p inside): f = f_0 * dr/|dr| * exp(-gap/l_0) points toward
+    # the center for interior points near the wall
+    mags = np.linalg.norm(f, axis=0)
+    assert mags[0] == 0.0  # node at origin: r_mag=0 -> u_hat undefined? node 0 at x=0
+    assert np.all(np.diff(mags[1:]) > 0)  # monotone growth toward the wall
+    # direction: -x (toward center) for points on the +x axis
+    assert np.all(f[0, 1:] < 0)
+    # clamped fiber skips node 0
+    fib.minus_clamped = True
+    f2 = fib.periphery_repulsion("sphere", radius=1.0)
+    assert np.all(f2[:, 0] == 0) and np.allclose(f2[:, 1:], f[:, 1:])
+    # ellipsoid variant: finite and inward on the x axis
+    f3 = fib.periphery_repulsion("ellipsoid", abc=(1.0, 0.8, 0.8))
+    assert np.isfinite(f3).all() and np.all(f3[0, 1:] < 0)
+
 
 @pytest.mark.timeout(300)
 def test_free_fiber_advects_with_uniform_flow():
