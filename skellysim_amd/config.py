@@ -1,0 +1,78 @@
+"""Reference TOML config consumption: reads the config files SkellySim's own
+generator (src/skelly_sim/skelly_config.py) produces — params, fibers (with
+the flat col-major x arrays), periphery — and builds a SystemFD (SURVEY.md §5
+"New build keeps the same TOML keys it consumes so reference configs run
+unchanged").
+
+The reference feeds the periphery matrices from its precompute .npz
+(precompute_file); this engine assembles those operators ON DEVICE
+(periphery_precompute.assemble_shell_operator) from the geometry
+(nodes/normals/weights), which build_system takes as an npz path or dict."""
+
+import numpy as np
+import tomli
+
+from .fiber_fd import FiberFD
+from .system_fd import SystemFD, Shell
+
+
+def load_config(path):
+    with open(path, "rb") as f:
+        return tomli.load(f)
+
+
+def build_fibers(cfg, eta):
+    fibers = []
+    for ft in cfg.get("fibers", []):
+        x = np.asarray(ft["x"], float).reshape(-1, 3)  # flat col-major 3 x n
+        fibers.append(FiberFD(
+            x, length=ft["length"], bending_rigidity=ft["bending_rigidity"],
+            eta=eta, radius=ft.get("radius", 0.0125),
+            force_scale=ft.get("force_scale", 0.0),
+            minus_clamped=bool(ft.get("minus_clamped", False))))
+    return fibers
+
+
+def periphery_interaction_from(cfg):
+    """fiber-periphery steric params (params.cpp:18,75-78; defaults
+    f_0=20, l_0=0.05, params.hpp:46-47) when the flag is on."""
+    p = cfg.get("params", {})
+    if not p.get("periphery_interaction_flag", False):
+        return None
+    per = cfg.get("periphery", {})
+    fp = p.get("fiber_periphery_interaction", {})
+    kw = dict(f_0=fp.get("f_0", 20.0), l_0=fp.get("l_0", 0.05))
+    if per.get("shape") == "sphere":
+        return dict(kind="sphere", radius=per["radius"], **kw)
+    if per.get("shape") == "ellipsoid":
+        return dict(kind="ellipsoid", abc=(per["a"], per["b"], per["c"]), **kw)
+    return None
+
+
+def build_system(cfg, backend=None, shell_geometry=None, dt=None,
+                 background_flow=None):
+    """SystemFD from a reference config. shell_geometry: npz path or dict
+    with nodes/normals/quadrature_weights (the periphery geometry the
+    reference's precompute generates; operators are assembled on device)."""
+    params = cfg.get("params", {})
+    eta = params.get("eta", 1.0)
+    dt = dt if dt is not None else params.get("dt_initial", 0.025)
+    fibers = build_fibers(cfg, eta)
+
+    shell = None
+    if "periphery" in cfg and shell_geometry is not None:
+        import torch
+        from .periphery_precompute import assemble_shell_operator
+
+        g = np.load(shell_geometry) if isinstance(shell_geometry, str) \
+            else shell_geometry
+        dev = torch.device("cuda:0")
+        A, M_inv = assemble_shell_operator(
+            torch.from_numpy(np.asarray(g["nodes"])).to(dev),
+            torch.from_numpy(np.asarray(g["normals"])).to(dev),
+            torch.from_numpy(np.asarray(g["quadrature_weights"])).to(dev))
+        shell = Shell(np.asarray(g["nodes"]), np.asarray(g["normals"]), A, M_inv)
+
+    return SystemFD(fibers, eta=eta, dt=dt, shell=shell,
+                    background_flow=background_flow, backend=backend,
+                    periphery_interaction=periphery_interaction_from(cfg))

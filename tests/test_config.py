@@ -1,0 +1,62 @@
+"""CPU tests: the reference-generator-authored config fixture loads and
+builds the config-4 system (fiber geometry consistent with the committed
+ellipsoid periphery geometry)."""
+
+import os
+
+import numpy as np
+
+from skellysim_amd.config import load_config, build_fibers, periphery_interaction_from
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+CFG = os.path.join(HERE, "golden", "skelly_config_ellipsoid.toml")
+
+
+def test_load_reference_config():
+    cfg = load_config(CFG)
+    assert cfg["params"]["eta"] == 1.0
+    assert cfg["params"]["fiber_type"] == "FiniteDifference"
+    assert cfg["periphery"]["shape"] == "ellipsoid"
+    assert len(cfg["fibers"]) == 512
+
+
+def test_build_fibers_geometry():
+    cfg = load_config(CFG)
+    fibers = build_fibers(cfg, eta=cfg["params"]["eta"])
+    assert len(fibers) == 512
+    assert all(f.n_nodes == 64 and f.minus_clamped for f in fibers)
+    a, b, c = (cfg["periphery"][k] for k in "abc")
+    for f in fibers[::64]:
+        # minus end sits just inside the attachment ellipsoid (a, b, c);
+        # the periphery node surface is a further x1.04 out (precompute.py:34)
+        p = f.x[:, 0]
+        lvl = (p[0] / a) ** 2 + (p[1] / b) ** 2 + (p[2] / c) ** 2
+        assert 0.8 < lvl < 1.0, lvl
+        # fiber points inward: second node strictly inside the first
+        q = f.x[:, 1]
+        lvl2 = (q[0] / a) ** 2 + (q[1] / b) ** 2 + (q[2] / c) ** 2
+        assert lvl2 < lvl
+        # fiber arclength == declared length
+        seg = np.diff(f.x.T, axis=0)
+        assert abs(np.linalg.norm(seg, axis=1).sum() - f.length) < 1e-2
+
+
+def test_fibers_inside_committed_periphery_geometry():
+    """The committed 8192-node geometry fixture (nodes at x1.04 the
+    attachment surface) encloses every fiber node."""
+    cfg = load_config(CFG)
+    g = np.load(os.path.join(HERE, "golden", "ellipsoid_8192_nodes.npz"))
+    a, b, c = float(g["a"]), float(g["b"]), float(g["c"])
+    fibers = build_fibers(cfg, eta=1.0)
+    for f in fibers:
+        lvl = ((f.x[0] / a) ** 2 + (f.x[1] / b) ** 2 + (f.x[2] / c) ** 2)
+        assert np.all(lvl < 1.0)
+
+
+def test_periphery_interaction_parsing():
+    cfg = load_config(CFG)
+    assert periphery_interaction_from(cfg) is None  # flag off in the example
+    cfg["params"]["periphery_interaction_flag"] = True
+    pi = periphery_interaction_from(cfg)
+    assert pi["kind"] == "ellipsoid" and pi["abc"] == (7.8, 4.16, 4.16)
+    assert pi["f_0"] == 20.0 and pi["l_0"] == 0.05
