@@ -449,3 +449,98 @@ class SystemFD:
         for f, a, b in self._fiber_slices():
             f.step(self.solution[a:b])
         return info
+
+    # ---- adaptive time-stepping driver (System::run, system.cpp:516-570) --
+    def fiber_error(self):
+        """max | |x_s| - 1 | over all fiber nodes
+        (FiberContainerFiniteDifference::fiber_error_local, f_c_fd.cpp:79-89)."""
+        err = 0.0
+        for f in self.fibers:
+            m = f.mats
+            xs = (2.0 / f.length) * f.x @ m["D_1_0"]
+            err = max(err, float(np.abs(np.linalg.norm(xs, axis=0) - 1.0).max()))
+        return err
+
+    def check_collision(self, periphery_shape=None, threshold=0.0):
+        """Fiber-periphery collision (f_c_fd.cpp:39-55 over
+        SphericalPeriphery::check_collision periphery.cpp:107-114 /
+        EllipsoidalPeriphery periphery.cpp:203-224); minus-clamped fibers
+        skip node 0. periphery_shape: dict like periphery_interaction."""
+        if periphery_shape is None:
+            return False
+        for f in self.fibers:
+            pc = f.x[:, 1:] if f.minus_clamped else f.x
+            if periphery_shape["kind"] == "sphere":
+                r2 = (periphery_shape["radius"] - threshold) ** 2
+                if np.any((pc ** 2).sum(axis=0) >= r2):
+                    return True
+            else:
+                a, b, c = periphery_shape["abc"]
+                r_s = pc / np.array([a, b, c])[:, None]
+                r_s_mag = np.linalg.norm(r_s, axis=0)
+                phi = np.arctan2(r_s[1], r_s[0] + 1e-12)
+                theta = np.arccos(r_s[2] / (1e-12 + r_s_mag))
+                st = np.sin(theta)
+                rc = np.stack([(a - threshold) * st * np.cos(phi),
+                               (b - threshold) * st * np.sin(phi),
+                               (c - threshold) * np.cos(theta)])
+                if np.any((pc ** 2).sum(axis=0) >= (rc ** 2).sum(axis=0)):
+                    return True
+        return False
+
+    def backup(self):
+        """System::backup (system.cpp:495-505): fiber state only here."""
+        self._bak = [(f.x.copy(), f.tension.copy()) for f in self.fibers]
+
+    def restore(self):
+        for f, (x, t) in zip(self.fibers, self._bak):
+            f.x = x
+            f.tension = t
+
+    def run(self, t_final, adaptive=True, dt_min=1e-4, dt_max=None,
+            beta_up=1.2, beta_down=0.5, fiber_error_tol=0.1,
+            periphery_shape=None, tol=1e-10, maxiter=300, restart=None,
+            on_accept=None):
+        """The reference timestep loop (System::run, system.cpp:516-570):
+        backup -> step -> accept if converged and fiber error within
+        tolerance (grow dt when comfortably inside, params.cpp:9-10 defaults
+        beta_up=1.2, beta_down=0.5), reject+restore+shrink otherwise; abort
+        below dt_min; collision rejects with dt/2. on_accept(system, time)
+        is the trajectory-write hook.
+
+        One deliberate deviation: the reference updates properties.dt to the
+        grown value BEFORE advancing time (system.cpp:553-561), so an
+        accepted step advances the clock by dt_new although the state moved
+        by the old dt; here time advances by the dt the step was actually
+        taken with."""
+        time_now = 0.0
+        dt_max = dt_max if dt_max is not None else self.dt
+        history = []
+        while time_now < t_final:
+            self.backup()
+            info = self.step(tol=tol, maxiter=maxiter, restart=restart)
+            err = self.fiber_error()
+            accept = True
+            dt_new = self.dt
+            if adaptive:
+                if info["converged"] and err <= fiber_error_tol:
+                    if err <= 0.9 * fiber_error_tol:
+                        dt_new = min(dt_max, self.dt * beta_up)
+                else:
+                    dt_new = self.dt * beta_down
+                    accept = False
+                if info["converged"] and self.check_collision(periphery_shape):
+                    dt_new = self.dt * 0.5
+                    accept = False
+                if dt_new < dt_min:
+                    raise RuntimeError("Timestep smaller than dt_min")
+            if accept:
+                time_now += self.dt
+                history.append(dict(time=time_now, dt=self.dt, iters=info["iters"],
+                                    fiber_error=err))
+                if on_accept is not None:
+                    on_accept(self, time_now)
+            else:
+                self.restore()
+            self.dt = dt_new
+        return history

@@ -163,6 +163,38 @@ def test_free_fiber_advects_with_uniform_flow():
 
 
 @pytest.mark.timeout(300)
+def test_adaptive_run_loop():
+    """System::run mirror: adaptive dt grows to dt_max on an easy problem
+    and the accepted-time bookkeeping reaches t_final exactly."""
+    U = np.array([0.1, 0.0, 0.0])
+    fib = straight_fiber(n=16)
+    x0 = fib.x.copy()
+    sys_ = SystemFD([fib], eta=1.0, dt=0.05, backend=OracleBackend(),
+                    background_flow=lambda r: np.tile(U, (len(r), 1)))
+    writes = []
+    hist = sys_.run(t_final=0.4, dt_max=0.1, tol=1e-12, maxiter=200,
+                    on_accept=lambda s, t: writes.append(t))
+    assert len(hist) >= 4
+    t_end = hist[-1]["time"]
+    assert t_end >= 0.4
+    assert hist[-1]["dt"] == pytest.approx(0.1)  # grew to dt_max
+    assert writes == [h["time"] for h in hist]
+    err = np.abs(fib.x - (x0 + U[:, None] * t_end)).max()
+    assert err < 1e-9, err
+
+
+@pytest.mark.timeout(300)
+def test_adaptive_run_rejects_below_dt_min():
+    """Unattainable fiber-error tolerance: every step rejected, dt shrinks,
+    run aborts below dt_min (system.cpp:548-551)."""
+    fib = straight_fiber(n=16)
+    sys_ = SystemFD([fib], eta=1.0, dt=0.05, backend=OracleBackend(),
+                    background_flow=lambda r: np.tile([0.1, 0, 0], (len(r), 1)))
+    with pytest.raises(RuntimeError, match="dt_min"):
+        sys_.run(t_final=1.0, fiber_error_tol=0.0, dt_min=1e-3, tol=1e-12)
+
+
+@pytest.mark.timeout(300)
 def test_two_fiber_system_converges_and_is_finite():
     """Two interacting fibers: GMRES converges and positions remain sane."""
     f1 = straight_fiber(n=32, direction=(0, 0, 1.0), x0=(0, 0, 0))
