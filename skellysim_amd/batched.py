@@ -34,9 +34,19 @@ def _lu_factor(mats):
 
 
 def robust_inv(A):
-    """torch.linalg.inv with the same magma fallback (hipblasDtrsm also hits
-    HIPBLAS_STATUS_ALLOC_FAILED at some sizes on ROCm 7.2)."""
+    """torch.linalg.inv with the magma backend forced for large device
+    matrices: ROCm 7.2's hipBLAS path hits HIPBLAS_STATUS_ALLOC_FAILED at
+    some sizes (hipblasDtrsm), and in long-running processes the same
+    workspace allocation can abort() outright (uncatchable) — measured on
+    MI355X at 24576^2 mid-pytest-session. magma handles every size used
+    here; cached allocator blocks are released first so magma's own
+    workspace allocation cannot hit the same wall."""
     global _magma_latched
+    if A.is_cuda and A.shape[-1] >= 4096:
+        if not _magma_latched:
+            torch.backends.cuda.preferred_linalg_library("magma")
+            _magma_latched = True
+        torch.cuda.empty_cache()
     try:
         return torch.linalg.inv(A)
     except RuntimeError as e:
