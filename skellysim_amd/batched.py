@@ -33,20 +33,45 @@ def _lu_factor(mats):
         return torch.linalg.lu_factor(mats)
 
 
+_huge_inv_done = False
+
+
+def _host_inv(A):
+    import scipy.linalg as scla
+    return torch.from_numpy(scla.inv(A.cpu().numpy())).to(A.device)
+
+
 def robust_inv(A):
-    """torch.linalg.inv with the magma backend forced for large device
-    matrices: ROCm 7.2's hipBLAS path hits HIPBLAS_STATUS_ALLOC_FAILED at
-    some sizes (hipblasDtrsm), and in long-running processes the same
-    workspace allocation can abort() outright (uncatchable) — measured on
-    MI355X at 24576^2 mid-pytest-session. magma handles every size used
-    here; cached allocator blocks are released first so magma's own
-    workspace allocation cannot hit the same wall."""
-    global _magma_latched
-    if A.is_cuda and A.shape[-1] >= 4096:
-        if not _magma_latched:
-            torch.backends.cuda.preferred_linalg_library("magma")
-            _magma_latched = True
-        torch.cuda.empty_cache()
+    """torch.linalg.inv hardened against ROCm 7.2 linalg failure modes
+    measured on MI355X (all at fp64):
+      - hipBLAS (default backend) inverts 24576^2 fine in a fresh process
+        but ABORTS (uncatchable) on a repeat in the same process, and hits
+        a catchable HIPBLAS_STATUS_ALLOC_FAILED at 18000^2;
+      - magma inverts 18000^2 fine but SEGFAULTS at 24576^2.
+    Strategy: release cached blocks first; for huge matrices (>= 20000)
+    force the default backend, allow one in-process GPU inversion and route
+    repeats through host LAPACK (~15 s once per geometry — setup-time);
+    below that, default backend with the magma fallback on the catchable
+    ALLOC failure."""
+    global _magma_latched, _huge_inv_done
+    if not A.is_cuda:
+        return torch.linalg.inv(A)
+    n = A.shape[-1]
+    torch.cuda.empty_cache()
+    if n >= 20000:
+        if _huge_inv_done:
+            return _host_inv(A)
+        if _magma_latched:
+            torch.backends.cuda.preferred_linalg_library("default")
+        try:
+            out = torch.linalg.inv(A)
+        except RuntimeError:
+            out = _host_inv(A)
+        finally:
+            if _magma_latched:
+                torch.backends.cuda.preferred_linalg_library("magma")
+        _huge_inv_done = True
+        return out
     try:
         return torch.linalg.inv(A)
     except RuntimeError as e:
