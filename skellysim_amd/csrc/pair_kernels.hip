@@ -32,6 +32,8 @@
 
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
+
 #define BLOCK 256
 #define TILE 512
 
@@ -378,10 +380,16 @@ hipError_t launch_pair(const double *r_src, const double *f_src, const double *r
     const long long blocks = (n_trg + (long long)BLOCK * tpt - 1) / ((long long)BLOCK * tpt);
 
     /* Source-split when the target grid alone underfills the chip (256 CUs;
-     * aim >= ~512 workgroups) and there are enough sources to slice. */
+     * aim >= SKELLY_SPLIT_TARGET workgroups, default 512) and there are
+     * enough sources to slice. */
+    static const long long split_target = [] {
+        const char *e = getenv("SKELLY_SPLIT_TARGET");
+        long long v = e ? atoll(e) : 512;
+        return v > 0 ? v : 512;
+    }();
     int n_slices = 1;
-    if (blocks < 512) {
-        long long s = (512 + blocks - 1) / blocks;
+    if (blocks < split_target) {
+        long long s = (split_target + blocks - 1) / blocks;
         long long max_by_src = (n_src + 2047) / 2048; /* keep >= ~2048 src/slice */
         n_slices = (int)(s < max_by_src ? s : max_by_src);
         if (n_slices < 1)
