@@ -29,10 +29,14 @@ def main():
     ap.add_argument("--fibers", type=int, default=4000)
     ap.add_argument("--nodes", type=int, default=32)
     ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--geometry", default="sphere", choices=["sphere", "oocyte"],
+                    help="oocyte = the reference example's surface of revolution "
+                         "(tests/golden/oocyte_nodes.npz)")
     args = ap.parse_args()
 
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    fix = np.load(os.path.join(repo, "tests", "golden", "sphere_6000_nodes.npz"))
+    geom = "sphere_6000_nodes.npz" if args.geometry == "sphere" else "oocyte_nodes.npz"
+    fix = np.load(os.path.join(repo, "tests", "golden", geom))
     dev = torch.device("cuda:0")
 
     t0 = time.perf_counter()
@@ -49,6 +53,7 @@ def main():
     for i in sel:
         p = fix["nodes"][i]
         n = fix["normals"][i] / np.linalg.norm(fix["normals"][i])
+        n = -n if np.dot(n, p) > 0 else n  # point inward
         s = np.linspace(0.05, 0.05 + length, args.nodes)
         fibers.append(FiberFD(p[None, :] + s[:, None] * n[None, :], length=length,
                               bending_rigidity=E, eta=1.0, minus_clamped=True,
@@ -70,7 +75,10 @@ def main():
     # velocity-field cross-check vs the CPU oracle on the FINAL state
     import oracle
     rng = np.random.default_rng(2)
-    pts = rng.uniform(-0.5, 0.5, (64, 3)) * float(fix["radius"])
+    if args.geometry == "sphere":
+        pts = rng.uniform(-0.5, 0.5, (64, 3)) * float(fix["radius"])
+    else:
+        pts = 0.35 * fix["nodes"][rng.integers(0, len(fix["nodes"]), 64)]
     r_fib = sys_.fiber_nodes()
     w = np.concatenate([f.quadrature_weights() for f in fibers])
     fw = np.zeros_like(r_fib)
