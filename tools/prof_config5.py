@@ -44,7 +44,9 @@ def main():
                                        torch.from_numpy(fix["normals"]).to(dev),
                                        torch.from_numpy(fix["quadrature_weights"]).to(dev))
     torch.cuda.synchronize()
-    print(f"shell operator (6000 nodes): {time.perf_counter()-t0:.2f}s")
+    
+    print(f"shell operator ({len(fix['nodes'])} nodes): "
+          f"{time.perf_counter()-t0:.2f}s")
     shell = Shell(fix["nodes"], fix["normals"], A, M_inv)
 
     length, E = 1.0, 2.5e-3
