@@ -50,16 +50,34 @@ def main():
     shell = Shell(fix["nodes"], fix["normals"], A, M_inv)
 
     length, E = 1.0, 2.5e-3
-    sel = np.linspace(0, len(fix["nodes"]) - 1, args.fibers).astype(int)
+
+    def inside(pts):
+        """All pts strictly inside the periphery (with margin)."""
+        if args.geometry == "sphere":
+            return np.all(np.linalg.norm(pts, axis=1) < float(fix["radius"]) - 0.02)
+        T, p1, p2, L = (float(fix[k]) for k in
+                        ("envelope_T", "envelope_p1", "envelope_p2", "envelope_length"))
+        q = pts / float(fix["scale_factor"])  # envelope is pre-scale
+        xq = np.clip(2 * q[:, 0] / L, -0.999, 0.999)
+        h = 0.5 * T * (1 + xq) ** p1 * (1 - xq) ** p2 * L
+        return np.all(q[:, 1] ** 2 + q[:, 2] ** 2 < (h - 0.02) ** 2) \
+            and np.all(np.abs(q[:, 0]) < L / 2 - 0.05)
+
     fibers = []
-    for i in sel:
+    order = np.random.default_rng(0).permutation(len(fix["nodes"]))
+    for i in order:
+        if len(fibers) == args.fibers:
+            break
         p = fix["nodes"][i]
         n = fix["normals"][i] / np.linalg.norm(fix["normals"][i])
         n = -n if np.dot(n, p) > 0 else n  # point inward
         s = np.linspace(0.05, 0.05 + length, args.nodes)
-        fibers.append(FiberFD(p[None, :] + s[:, None] * n[None, :], length=length,
-                              bending_rigidity=E, eta=1.0, minus_clamped=True,
-                              force_scale=-0.05))
+        x = p[None, :] + s[:, None] * n[None, :]
+        if not inside(x):
+            continue  # tip regions: an inward fiber would pierce the far wall
+        fibers.append(FiberFD(x, length=length, bending_rigidity=E, eta=1.0,
+                              minus_clamped=True, force_scale=-0.05))
+    print(f"placed {len(fibers)} fibers")
 
     sys_ = SystemFD(fibers, eta=1.0, dt=0.025, shell=shell, backend=HipBackend())
     print(f"solution size: {sys_.fiber_sol_size + sys_.shell_sol_size}")
