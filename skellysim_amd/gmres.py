@@ -143,8 +143,14 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
         else:
             break
 
-    # final check against the true residual
+    # final check against the TRUE residual: the implicit (Givens) residual
+    # can report convergence the operator does not support, e.g. when a
+    # numerically broken preconditioner makes A M^-1 inconsistent
+    true_resid = None
     if converged:
         true_resid = float(norm(b - matvec(x)) / bnorm)
         residuals.append(true_resid)
-    return x, {"converged": converged, "iters": total_iters, "residuals": residuals}
+        if true_resid > max(100.0 * tol, 1e-12):
+            converged = False
+    return x, {"converged": converged, "iters": total_iters,
+               "residuals": residuals, "true_residual": true_resid}
