@@ -185,8 +185,7 @@ class SystemFD:
 
     # ---- solver pipeline ------------------------------------------------
     def prep_state_for_solver(self):
-        """system.cpp:396-459 (no bodies, no dynamic instability, no
-        fiber-periphery repulsion)."""
+        """system.cpp:396-459 (no bodies, no dynamic instability)."""
         dt, eta = self.dt, self.eta
         for f in self.fibers:
             f.update_constants(eta)
@@ -319,7 +318,6 @@ class SystemFD:
         (matvec + preconditioner) runs on device with zero host traffic —
         SURVEY.md §8f row 3 in full."""
         import torch
-        t = self.backend.torch
         dev = self.backend.dev
         T = lambda a: torch.from_numpy(np.ascontiguousarray(a)).to(dev)
 
