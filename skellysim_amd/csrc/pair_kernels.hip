@@ -591,20 +591,23 @@ hipError_t launch_oseen_tensor_batched(const double *pts, double *G, long long n
 /* ---- fp64 FMA peak microbenchmark (roofline denominator) -------------- */
 
 __global__ __launch_bounds__(256) void fp64_fma_peak_kernel(double *out, int iters) {
-    double a0 = 1.0 + 1e-9 * threadIdx.x, a1 = a0 + 0.1, a2 = a0 + 0.2, a3 = a0 + 0.3;
-    double a4 = a0 + 0.4, a5 = a0 + 0.5, a6 = a0 + 0.6, a7 = a0 + 0.7;
+    /* 16 independent accumulators: with 8, the chain is fp64-FMA
+     * dependent-latency-bound (~20% below issue rate) and under-reports the
+     * sustained peak. */
+    double a[16];
+#pragma unroll
+    for (int k = 0; k < 16; ++k)
+        a[k] = 1.0 + 1e-9 * threadIdx.x + 0.1 * k;
     const double b = 1.0 + 1e-12, c = 1e-12;
     for (int i = 0; i < iters; ++i) {
-        a0 = __builtin_fma(a0, b, c);
-        a1 = __builtin_fma(a1, b, c);
-        a2 = __builtin_fma(a2, b, c);
-        a3 = __builtin_fma(a3, b, c);
-        a4 = __builtin_fma(a4, b, c);
-        a5 = __builtin_fma(a5, b, c);
-        a6 = __builtin_fma(a6, b, c);
-        a7 = __builtin_fma(a7, b, c);
+#pragma unroll
+        for (int k = 0; k < 16; ++k)
+            a[k] = __builtin_fma(a[k], b, c);
     }
-    const double s = a0 + a1 + a2 + a3 + a4 + a5 + a6 + a7;
+    double s = 0.0;
+#pragma unroll
+    for (int k = 0; k < 16; ++k)
+        s += a[k];
     if (s == -1.0) /* never true; defeats DCE without a store per thread */
         out[blockIdx.x] = s;
 }
@@ -631,7 +634,7 @@ hipError_t run_fp64_peak(double *out_tflops) {
     (void)hipFree(d);
     if (err != hipSuccess)
         return err;
-    const double flops = 2.0 * 8.0 * (double)iters * (double)blocks * threads;
+    const double flops = 2.0 * 16.0 * (double)iters * (double)blocks * threads;
     *out_tflops = flops / (ms * 1e-3) / 1e12;
     return hipSuccess;
 }
