@@ -44,6 +44,32 @@ def test_n1e6_properties(ska):
     assert torch.equal(u_f, u_f2)
 
 
+def test_n1e6_stresslet_oseen_properties(ska):
+    """Full-size properties for the other two hot kernels: linearity in the
+    source strengths and eta scaling at N=1e6."""
+    rng = np.random.default_rng(100)
+    n = 1_000_000
+    dev = torch.device("cuda:0")
+    pts = torch.from_numpy(rng.uniform(-1, 1, (n, 3))).to(dev)
+    f9 = torch.from_numpy(rng.uniform(-1, 1, (n, 9))).to(dev)
+    g9 = torch.flip(f9, dims=[0]).contiguous()
+    u_f = ska.stresslet_device(pts, f9, pts, 1.0)
+    u_g = ska.stresslet_device(pts, g9, pts, 1.0)
+    u_s = ska.stresslet_device(pts, (f9 + g9).contiguous(), pts, 1.0)
+    torch.cuda.synchronize()
+    assert float(torch.norm(u_s - (u_f + u_g)) / torch.norm(u_s)) < 1e-12
+    u_eta = ska.stresslet_device(pts, f9, pts, 4.0)
+    torch.cuda.synchronize()
+    assert float(torch.norm(u_f - 4.0 * u_eta) / torch.norm(u_f)) < 1e-13
+
+    rho = torch.from_numpy(rng.uniform(-1, 1, (n, 3))).to(dev)
+    u1 = ska.oseen_contract_device(pts, pts, rho, 1.0)
+    u2 = ska.oseen_contract_device(pts, pts, rho, 2.0)
+    torch.cuda.synchronize()
+    assert float(torch.norm(u1 - 2.0 * u2) / torch.norm(u1)) < 1e-13
+    assert torch.isfinite(u1).all()
+
+
 def test_n1e6_subset_parity(ska, oracle_mod):
     """Direct oracle parity on a 256-target subset of the N=1e6 cloud
     (2.56e8 pairs on the host cores — seconds)."""

@@ -147,6 +147,66 @@ def test_streamline_request(tmp_path):
     assert np.median(relerr) < 0.25
 
 
+def test_velocity_field_with_shell(tmp_path):
+    """Shell branch of listener velocity_field: the frame's solution_vec_
+    drives the double-layer term (system.cpp:330-384, shell_->flow of the
+    stored density) on top of the fiber stokeslet flow."""
+    import os
+    from skellysim_amd.system_fd import Shell
+    here = os.path.dirname(os.path.abspath(__file__))
+    fx = np.load(os.path.join(here, "golden", "periphery_sphere_192.npz"))
+    shell = Shell(fx["nodes"], fx["normals"], fx["stresslet_plus_complementary"],
+                  fx["M_inv"])
+    s = np.linspace(0, 1.0, 16)
+    x = np.stack([0.1 * np.sin(np.pi * s), np.zeros_like(s), 0.6 * s - 0.8],
+                 axis=1)
+    fib = FiberFD(x, length=1.0, bending_rigidity=2.5e-2, eta=1.0,
+                  force_scale=-0.02)
+    sys_ = SystemFD([fib], eta=1.0, dt=0.05, shell=shell,
+                    backend=OracleBackend())
+    path = str(tmp_path / "skelly_sim.out")
+    with TrajectoryWriter(path) as tw:
+        assert sys_.step(tol=1e-11, maxiter=300)["converged"]
+        tw.write_frame(sys_, 0.05, 0.05)
+    dens = sys_.solution[sys_.fiber_sol_size:].reshape(-1, 3)
+    assert np.linalg.norm(dens) > 0  # the shell really participates
+
+    targets = np.array([[0.3, 0.1, -0.2], [0.0, 0.4, 0.5], [-0.5, 0.0, 0.0]])
+    traj = Trajectory(path)
+    stdin = io.BytesIO(_request_bytes(0, targets) + struct.pack("<Q", 0))
+    stdout = io.BytesIO()
+    serve(stdin, stdout, traj, OracleBackend(), eta=1.0,
+          shell_geometry={"nodes": fx["nodes"], "normals": fx["normals"]})
+    stdout.seek(0)
+    (size,) = struct.unpack("<Q", stdout.read(8))
+    res = msgpack.unpackb(stdout.read(size), raw=False)
+    u = eigen_decode(res["velocity_field"])
+
+    import oracle
+    f = sys_.fibers[0]
+    f.update_derivatives()        # the listener rebuilds operators from the
+    f.update_force_operator()     # frame's (post-step) positions
+    sol = np.concatenate([f.x.reshape(-1), f.tension])
+    ff = f.force_operator @ sol
+    fn = np.stack([ff[i * f.n_nodes:(i + 1) * f.n_nodes] for i in range(3)],
+                  axis=1)
+    ref = oracle.stokeslet(f.x.T, fn * f.quadrature_weights()[:, None],
+                           targets, 1.0)
+    f_dl = 2.0 * np.einsum("ni,nj->nij", fx["normals"], dens).reshape(-1, 9)
+    ref += oracle.stresslet(fx["nodes"], f_dl, targets, 1.0)
+    assert np.allclose(u, ref, atol=1e-12)
+
+    # without shell_geometry the shell term is (documented) absent
+    stdin = io.BytesIO(_request_bytes(0, targets) + struct.pack("<Q", 0))
+    stdout2 = io.BytesIO()
+    serve(stdin, stdout2, traj, OracleBackend(), eta=1.0)
+    stdout2.seek(0)
+    (size,) = struct.unpack("<Q", stdout2.read(8))
+    res2 = msgpack.unpackb(stdout2.read(size), raw=False)
+    u2 = eigen_decode(res2["velocity_field"])
+    assert not np.allclose(u2, ref, atol=1e-12)
+
+
 def test_invalid_frame_gives_empty_response(tmp_path):
     path, _ = _write_traj(tmp_path)
     reqs = [_request_bytes(99, np.zeros((1, 3))), _request_bytes(0, np.zeros((1, 3)))]
