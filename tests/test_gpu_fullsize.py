@@ -69,6 +69,12 @@ def test_n1e6_stresslet_oseen_properties(ska):
     assert float(torch.norm(u1 - 2.0 * u2) / torch.norm(u1)) < 1e-13
     assert torch.isfinite(u1).all()
 
+    w1 = ska.rotlet_device(pts, pts, rho, 1.0)
+    w2 = ska.rotlet_device(pts, pts, (2.0 * rho).contiguous(), 4.0)
+    torch.cuda.synchronize()
+    assert float(torch.norm(w1 - 2.0 * w2) / torch.norm(w1)) < 1e-13
+    assert torch.isfinite(w1).all()
+
 
 def test_n1e6_subset_parity(ska, oracle_mod):
     """Direct oracle parity on a 256-target subset of the N=1e6 cloud
