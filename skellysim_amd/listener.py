@@ -16,8 +16,10 @@ force_operator @ fiber_solution (apply_fiber_force on the frame's
 positions+tension), fiber flow WITHOUT self-subtraction (flow(..., false),
 system.cpp:355), plus the shell double layer from the frame's
 solution_vec_. Streamlines are integrated per the reference's adaptive
-5(4) RK scheme (see integrate_streamline); vortex lines (curl-field lines)
-are not implemented (empty list)."""
+5(4) RK scheme (see integrate_streamline); vortex lines are streamlines of
+the central-difference curl of the velocity field (streamline.cpp:16-35,
+115-165), with the singularity stop still on the VELOCITY norm
+(streamline.cpp:51, same observer for both line kinds)."""
 
 import argparse
 import struct
@@ -117,8 +119,27 @@ def velocity_field(frame, targets, eta, compute, shell_geometry=None):
     return u
 
 
+def vorticity(field_fn, pts, eps=1e-7):
+    """Central-difference curl of the velocity field at each row of pts
+    (get_vorticity_at_point, streamline.cpp:16-35: eps=1e-7, 6 evals/point;
+    0.5/eps times the +eps/-eps differences = the standard central-difference
+    curl)."""
+    pts = np.asarray(pts, float).reshape(-1, 3)
+    n = len(pts)
+    probes = np.repeat(pts, 6, axis=0)
+    for k in range(3):
+        probes[2 * k::6, k] += eps
+        probes[2 * k + 1::6, k] -= eps
+    v = field_fn(probes).reshape(n, 6, 3)
+    return 0.5 / eps * np.stack([
+        (v[:, 2, 2] - v[:, 3, 2]) - (v[:, 4, 1] - v[:, 5, 1]),
+        (v[:, 4, 0] - v[:, 5, 0]) - (v[:, 0, 2] - v[:, 1, 2]),
+        (v[:, 0, 1] - v[:, 1, 1]) - (v[:, 2, 0] - v[:, 3, 0]),
+    ], axis=1)
+
+
 def integrate_streamline(field_fn, x0, dt_init=0.1, t_final=1.0, abs_err=1e-10,
-                         rel_err=1e-6, back_integrate=True):
+                         rel_err=1e-6, back_integrate=True, rhs_fn=None):
     """One streamline (StreamLine::compute, src/core/streamline.cpp:66-118):
     adaptive 5(4) Runge-Kutta integration of dx/dt = u(x) forward to t_final
     (and backward when requested), with the reference's |u| > 1e3 singularity
@@ -126,15 +147,22 @@ def integrate_streamline(field_fn, x0, dt_init=0.1, t_final=1.0, abs_err=1e-10,
     (Dormand-Prince 5(4)) integrates the same ODE to the same tolerances —
     step placement differs, the curve does not.
 
-    Returns dict {x: (npts, 3), val: (npts, 3), time: [..]} (streamline.hpp:29).
-    """
+    rhs_fn overrides the integrated field (vortex lines integrate the curl,
+    VortexLine::compute streamline.cpp:115-165) while the singularity stop
+    stays on the VELOCITY field_fn (streamline.cpp:51).
+
+    Returns dict {x: (npts, 3), val: (npts, 3), time: [..]} (streamline.hpp:29),
+    val being the integrated field at the path points (velocity for
+    streamlines, vorticity for vortex lines)."""
     from scipy.integrate import solve_ivp
 
+    rfn = rhs_fn if rhs_fn is not None else field_fn
+
     def rhs(t, x):
-        return field_fn(x.reshape(1, 3)).reshape(3)
+        return rfn(x.reshape(1, 3)).reshape(3)
 
     def singularity(t, x):
-        return 1e3 - np.linalg.norm(rhs(t, x))
+        return 1e3 - np.linalg.norm(field_fn(x.reshape(1, 3)))
 
     singularity.terminal = True
 
@@ -153,17 +181,20 @@ def integrate_streamline(field_fn, x0, dt_init=0.1, t_final=1.0, abs_err=1e-10,
         x = np.concatenate([x_back[:0:-1], x_fwd])
     else:
         t, x = t_fwd, x_fwd
-    val = np.stack([field_fn(p.reshape(1, 3)).reshape(3) for p in x])
+    val = np.stack([rfn(p.reshape(1, 3)).reshape(3) for p in x])
     return {"x": x, "val": val, "time": t.tolist()}
 
 
-def process_streamlines(frame, req, eta, compute, shell_geometry):
+def process_streamlines(frame, req, eta, compute, shell_geometry, vortex=False):
+    """process_streamlines / process_vortexlines (listener.cpp:51-74): one
+    line per seed column; vortex=True integrates the curl field instead."""
     req = req or {}
     x0 = eigen_decode(req.get("x0", []))
     x0 = np.asarray(x0, float).reshape(-1, 3) if np.size(x0) else np.zeros((0, 3))
     if not len(x0):
         return []
     field = lambda pts: velocity_field(frame, pts, eta, compute, shell_geometry)
+    rhs_fn = (lambda pts: vorticity(field, pts)) if vortex else None
     out = []
     for seed in x0:
         s = integrate_streamline(field, seed,
@@ -171,7 +202,8 @@ def process_streamlines(frame, req, eta, compute, shell_geometry):
                                  t_final=float(req.get("t_final", 1.0)),
                                  abs_err=float(req.get("abs_err", 1e-10)),
                                  rel_err=float(req.get("rel_err", 1e-6)),
-                                 back_integrate=bool(req.get("back_integrate", True)))
+                                 back_integrate=bool(req.get("back_integrate", True)),
+                                 rhs_fn=rhs_fn)
         out.append({"x": eigen_encode_3xn(s["x"]),
                     "val": eigen_encode_3xn(s["val"]),
                     "time": s["time"]})
@@ -211,7 +243,9 @@ def serve(stdin, stdout, traj, compute, eta=1.0, shell_geometry=None):
             "n_frames": len(traj),
             "streamlines": process_streamlines(frame, cmd.get("streamlines"),
                                                eta, compute, shell_geometry),
-            "vortexlines": [],  # curl-field lines: out of scope this round
+            "vortexlines": process_streamlines(frame, cmd.get("vortexlines"),
+                                               eta, compute, shell_geometry,
+                                               vortex=True),
             "velocity_field": eigen_encode_3xn(u),
         }
         out = msgpack.packb(response)

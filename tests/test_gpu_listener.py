@@ -38,3 +38,45 @@ def test_listener_velocity_field_hip(tmp_path, hip_lib_path):
     ref = oracle.stokeslet(np.concatenate(r_src), np.concatenate(wf), targets, 1.0)
     rel = np.linalg.norm(u - ref) / np.linalg.norm(ref)
     assert rel < 1e-10, rel
+
+
+def test_listener_vortexline_hip(tmp_path, hip_lib_path):
+    """Vortex-line request served by the HIP compute path: val equals the
+    oracle-field central-difference vorticity at the returned points (the
+    1e-14 field parity is amplified by the eps=1e-7 differencing, hence the
+    loose tolerance)."""
+    from skellysim_amd.fiber_fd import FiberFD
+    from skellysim_amd.system_fd import SystemFD, HipBackend
+    from skellysim_amd.trajectory import TrajectoryWriter
+    from skellysim_amd.listener import (Trajectory, serve, eigen_decode,
+                                        velocity_field, vorticity)
+    from test_listener import _ndencode, _roundtrip
+    from oracle_backend import OracleBackend
+
+    s = np.linspace(0, 1.0, 24)
+    x = np.stack([0.15 * np.sin(2 * np.pi * s), np.zeros_like(s), s], axis=1)
+    fib = FiberFD(x, length=1.0, bending_rigidity=2.5e-2, eta=1.0)
+    sys_ = SystemFD([fib], eta=1.0, dt=0.1, backend=HipBackend())
+    path = str(tmp_path / "skelly_sim.out")
+    with TrajectoryWriter(path) as tw:
+        tw.write_frame(sys_, 0.1, 0.1)
+    cmd = {
+        "frame_no": 0,
+        "evaluator": "GPU",
+        "streamlines": {"dt_init": 0.1, "t_final": 1.0, "abs_err": 1e-10,
+                        "rel_err": 1e-6, "back_integrate": True,
+                        "x0": np.zeros((0, 3))},
+        "vortexlines": {"dt_init": 0.05, "t_final": 0.2, "abs_err": 1e-10,
+                        "rel_err": 1e-8, "back_integrate": False,
+                        "x0": np.array([[0.6, 0.2, 0.4]])},
+        "velocity_field": {"x": np.zeros((0, 3))},
+    }
+    msg = msgpack.packb(cmd, default=_ndencode)
+    (res,) = _roundtrip(path, [struct.pack("<Q", len(msg)) + msg], HipBackend())
+    (vl,) = res["vortexlines"]
+    xp = eigen_decode(vl["x"])
+    val = eigen_decode(vl["val"])
+    traj = Trajectory(path)
+    field = lambda p: velocity_field(traj.frames[0], p, 1.0, OracleBackend())
+    w = vorticity(field, xp)
+    assert np.linalg.norm(w - val) / np.linalg.norm(w) < 1e-4

@@ -147,6 +147,79 @@ def test_streamline_request(tmp_path):
     assert np.median(relerr) < 0.25
 
 
+def test_vorticity_helper():
+    """vorticity() = central-difference curl (streamline.cpp:16-35): exact
+    on rigid rotation (curl = 2*Omega), ~1e-8 on the analytic Stokeslet
+    vorticity f x r / (4 pi eta r^3)."""
+    from skellysim_amd.listener import vorticity
+    Om = np.array([0.3, -0.7, 0.5])
+    rot = lambda p: np.cross(np.tile(Om, (len(p), 1)), p)
+    pts = np.array([[0.2, 0.1, -0.3], [1.0, 2.0, 0.5]])
+    w = vorticity(rot, pts)
+    assert np.allclose(w, 2 * Om, atol=1e-7)
+
+    import oracle
+    src = np.array([[0.1, -0.2, 0.3]])
+    f = np.array([[0.7, 0.4, -0.9]])
+    eta = 1.3
+    trg = np.array([[1.5, 2.1, 1.7], [-1.8, 0.9, 2.2]])
+    field = lambda p: oracle.stokeslet(src, f, p, eta)
+    w = vorticity(field, trg)
+    r = trg - src
+    wa = np.cross(np.tile(f, (len(trg), 1)), r) \
+        / (4 * np.pi * eta * np.linalg.norm(r, axis=1)[:, None] ** 3)
+    assert np.linalg.norm(w - wa) / np.linalg.norm(wa) < 1e-6
+
+
+def test_vortexline_request(tmp_path):
+    """Vortex lines: seeds integrate along the CURL of the frame's velocity
+    field (VortexLine::compute, streamline.cpp:115-165) — path tangents match
+    the returned vorticities, and val recomputes as vorticity()."""
+    from skellysim_amd.listener import vorticity, velocity_field
+    s = np.linspace(0, 1.0, 24)
+    x = np.stack([0.15 * np.sin(2 * np.pi * s), np.zeros_like(s), s], axis=1)
+    fib = FiberFD(x, length=1.0, bending_rigidity=2.5e-2, eta=1.0)
+    sys_ = SystemFD([fib], eta=1.0, dt=0.1, backend=OracleBackend())
+    path = str(tmp_path / "skelly_sim.out")
+    with TrajectoryWriter(path) as tw:
+        tw.write_frame(sys_, 0.1, 0.1)
+    cmd = {
+        "frame_no": 0,
+        "evaluator": "GPU",
+        "streamlines": {"dt_init": 0.1, "t_final": 1.0, "abs_err": 1e-10,
+                        "rel_err": 1e-6, "back_integrate": True,
+                        "x0": np.zeros((0, 3))},
+        "vortexlines": {"dt_init": 0.05, "t_final": 0.3, "abs_err": 1e-10,
+                        "rel_err": 1e-8, "back_integrate": True,
+                        "x0": np.array([[0.6, 0.2, 0.4]])},
+        "velocity_field": {"x": np.zeros((0, 3))},
+    }
+    msg = msgpack.packb(cmd, default=_ndencode)
+    (res,) = _roundtrip(path, [struct.pack("<Q", len(msg)) + msg], OracleBackend())
+    assert res["streamlines"] == []
+    assert len(res["vortexlines"]) == 1
+    vl = res["vortexlines"][0]
+    x = eigen_decode(vl["x"])
+    val = eigen_decode(vl["val"])
+    t = np.asarray(vl["time"])
+    assert x.shape == val.shape and x.shape[1] == 3 and len(t) == len(x)
+    assert np.all(np.diff(t) > 0) and t[0] < 0 < t[-1]
+    i0 = np.argmin(np.abs(t))
+    assert np.allclose(x[i0], [0.6, 0.2, 0.4], atol=1e-12)
+    # val really is the vorticity of the frame's field at the path points
+    traj = Trajectory(path)
+    field = lambda p: velocity_field(traj.frames[0], p, 1.0, OracleBackend())
+    w = vorticity(field, x)
+    assert np.linalg.norm(w - val) / np.linalg.norm(w) < 1e-5
+    # path tangents track the vorticity (the integrated field)
+    dxdt = (x[2:] - x[:-2]) / (t[2:] - t[:-2])[:, None]
+    mid = val[1:-1]
+    mask = np.linalg.norm(mid, axis=1) > 1e-8
+    relerr = np.linalg.norm(dxdt[mask] - mid[mask], axis=1) / \
+        np.linalg.norm(mid[mask], axis=1)
+    assert np.median(relerr) < 0.25
+
+
 def test_velocity_field_with_shell(tmp_path):
     """Shell branch of listener velocity_field: the frame's solution_vec_
     drives the double-layer term (system.cpp:330-384, shell_->flow of the
