@@ -11,8 +11,16 @@ own target block. GMRES runs with distributed inner products
 
 The reference hard-forbids its direct evaluators under >1 rank
 (system.cpp:618-623); this module is the new multi-GPU capability.
-Covered by gloo world-2 CPU tests (tests/test_dist_system.py) with the
-oracle backend; the same code drives RCCL + HIP kernels on multi-GPU boxes.
+
+Two iteration paths share the math: a host (numpy) path, and the
+device-resident path (_apply_matvec_device/_apply_precond_device) where the
+whole GMRES iteration — batched fiber blocks, pair kernels, row-block shell
+GEMVs, distributed dot-blocks — runs on device tensors with only the wf /
+shell-density all-gathers per iteration (RCCL over xGMI on GPUs) plus one
+fiber-position gather per solve. Covered by gloo world-2 CPU tests
+(tests/test_dist_system.py; the device layout runs there on CPU tensors with
+oracle-backed kernel fakes) and by a world-1 HIP test (test_gpu_system.py);
+the same code drives RCCL + HIP kernels on multi-GPU boxes.
 """
 
 import numpy as np
@@ -155,16 +163,105 @@ class DistributedSystemFD(SystemFD):
             res[self.fiber_sol_size:] = self._shell_precond_rows(x_shell_all)
         return res
 
-    def solve(self, tol=1e-10, maxiter=200, restart=None):
-        """Distributed GMRES over the rank-local slices."""
+    # ---- device-resident distributed iteration --------------------------
+    @staticmethod
+    def _gather_t(t):
+        """allgather_rows when a world exists, identity otherwise (tensor)."""
+        import torch.distributed as dist
+        if dist.is_available() and dist.is_initialized() \
+                and dist.get_world_size() > 1:
+            return allgather_rows(t)
+        return t
+
+    def _build_device_operators(self):
+        """The single-rank builder works unchanged on the rank-local state
+        (all_nodes() is the local override, shell.A/M_inv are the row
+        blocks, shell.nodes/normals the global geometry); add the one
+        solve-static gather: fiber source positions from all ranks."""
+        super()._build_device_operators()
+        self._dev["r_fib_all"] = self._gather_t(self._dev["r_fib"])
+
+    def _apply_matvec_device(self, x):
+        """apply_matvec on device with all-gathered sources, local targets
+        (the host version above, collective for collective: wf + shell
+        density per iteration over RCCL/xGMI; positions once per solve)."""
+        from .evaluator import stokeslet_device, stresslet_device
+        d = self._dev
+        nf, n = d["nf"], d["n"]
+        eta = self.eta
+        nf_nodes = nf * n  # LOCAL fiber nodes (targets); sources are global
+        x_fib = x[: 4 * nf_nodes].reshape(nf, 4 * n)
+        x_shell_local = x[4 * nf_nodes:]
+
+        fw = torch.bmm(d["F"], x_fib.unsqueeze(-1)).squeeze(-1)      # (nf, 3n)
+        fw_nodes = fw.reshape(nf, 3, n).permute(0, 2, 1).reshape(nf_nodes, 3)
+        wf = (fw_nodes * d["w"][:, None]).contiguous()
+        wf_all = self._gather_t(wf)
+
+        v_all = stokeslet_device(d["r_fib_all"], wf_all, d["r_all"], eta)
+        # self subtraction uses the OWN fibers' wf only (f_c_fd.cpp:203-210)
+        corr = torch.bmm(d["G"], wf.reshape(nf, 3 * n, 1)).reshape(nf_nodes, 3)
+        v_all[:nf_nodes] -= corr
+
+        dens_all = None
+        if self.shell:
+            dens_all = self._gather_t(x_shell_local.reshape(-1, 3).contiguous())
+            f_dl = 2.0 * eta * torch.einsum("ni,nj->nij", d["sh_normals"],
+                                            dens_all).reshape(-1, 9).contiguous()
+            v_all[:nf_nodes] += stresslet_device(d["sh_nodes"], f_dl,
+                                                 d["r_fib"], eta)
+
+        res = torch.empty_like(x)
+        res[: 4 * nf_nodes] = self._fiber_block_device(
+            x_fib, v_all[:nf_nodes]).reshape(-1)
+        if self.shell:
+            # row-block GEMV against the GLOBAL density (periphery.cpp:34-47)
+            res[4 * nf_nodes:] = d["sh_A"] @ dens_all.reshape(-1) \
+                + v_all[nf_nodes:].reshape(-1)
+        return res
+
+    def _apply_precond_device(self, x):
+        d = self._dev
+        nf, n = d["nf"], d["n"]
+        res = torch.empty_like(x)
+        res[: 4 * nf * n] = d["lu"].solve(
+            x[: 4 * nf * n].reshape(nf, 4 * n)).reshape(-1)
+        if self.shell:
+            x_sh_all = self._gather_t(
+                x[4 * nf * n:].reshape(-1, 3).contiguous()).reshape(-1)
+            res[4 * nf * n:] = d["sh_Minv"] @ x_sh_all
+        return res
+
+    def solve(self, tol=1e-10, maxiter=200, restart=None, device_mode=None):
+        """Distributed GMRES over the rank-local slices. device_mode=True
+        runs the whole iteration on device tensors (matvec/precond above,
+        distributed dot-blocks in gmres); default auto-detects the product
+        backend. Requires >= 1 fiber per rank with uniform discretization."""
         import torch.distributed as dist
         from .gmres import gmres
+        from .system_fd import HipBackend
 
         rhs = self.prep_state_for_solver()
         if restart is None:
             restart = min(200, maxiter)
         distributed = dist.is_available() and dist.is_initialized() \
             and dist.get_world_size() > 1
+        if device_mode is None:
+            device_mode = bool(self.fibers) and self._uniform \
+                and isinstance(self.backend, HipBackend)
+
+        if device_mode:
+            self._build_device_operators()
+            b = self.backend._t(rhs)
+            x, info = gmres(self._apply_matvec_device, b,
+                            precond=self._apply_precond_device,
+                            tol=tol, maxiter=maxiter, restart=restart,
+                            distributed=distributed)
+            if b.is_cuda:
+                self.backend.torch.cuda.synchronize()
+            self.solution = x.cpu().numpy()
+            return info
+
         b = torch.from_numpy(rhs)
         mv = lambda v: torch.from_numpy(self.apply_matvec(v.numpy()))
         pc = lambda v: torch.from_numpy(self.apply_preconditioner(v.numpy()))

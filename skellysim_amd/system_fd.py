@@ -379,9 +379,22 @@ class SystemFD:
             v_all[:nf_nodes] += stresslet_device(d["sh_nodes"], f_dl,
                                                  d["r_fib"], eta)
 
-        # fiber matvec: A x - vT_in + xs_vT (fiber_fd.matvec, cpp:278-315)
-        v_fib = v_all[:nf_nodes].reshape(nf, n, 3).permute(0, 2, 1)  # (nf, 3, n)
-        vT = torch.zeros((nf, 4 * n), dtype=x.dtype, device=x.device)
+        res = torch.empty_like(x)
+        res[: 4 * nf_nodes] = self._fiber_block_device(x_fib, v_all[:nf_nodes]).reshape(-1)
+        if self.shell:
+            v_shell = v_all[nf_nodes:].reshape(-1)
+            res[4 * nf_nodes:] = d["sh_A"] @ x_shell + v_shell
+        return res
+
+    def _fiber_block_device(self, x_fib, v_fib_nodes):
+        """Per-fiber operator block: A x - vT_in + the two BC velocity
+        corrections (fiber_fd.matvec, fiber_finite_difference.cpp:278-315).
+        x_fib (nf, 4n), v_fib_nodes (nf*n, 3) node-major -> (nf, 4n)."""
+        import torch
+        d = self._dev
+        nf, n = d["nf"], d["n"]
+        v_fib = v_fib_nodes.reshape(nf, n, 3).permute(0, 2, 1)       # (nf, 3, n)
+        vT = torch.zeros((nf, 4 * n), dtype=x_fib.dtype, device=x_fib.device)
         vT[:, : 3 * n] = v_fib.reshape(nf, 3 * n)
         tens = torch.einsum("ts,fis->fit", d["D1preT"],
                             d["xs"] * v_fib).sum(dim=1)
@@ -393,13 +406,7 @@ class SystemFD:
         res_fib[:, bc_start + 3] += (v_fib[:, :, 0] * d["xs"][:, :, 0]).sum(dim=1)
         res_fib[:, bc_start + 10] += d["plus_vel"] * \
             (v_fib[:, :, -1] * d["xs"][:, :, -1]).sum(dim=1)
-
-        res = torch.empty_like(x)
-        res[: 4 * nf_nodes] = res_fib.reshape(-1)
-        if self.shell:
-            v_shell = v_all[nf_nodes:].reshape(-1)
-            res[4 * nf_nodes:] = d["sh_A"] @ x_shell + v_shell
-        return res
+        return res_fib
 
     def _apply_precond_device(self, x):
         import torch

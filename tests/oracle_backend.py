@@ -38,3 +38,42 @@ class OracleBackend:
         return (lambda x: A @ x), (lambda x: M_inv @ x)
 
 
+class FakeDeviceBackend(OracleBackend):
+    """TEST-ONLY: drives the device-resident code paths on CPU torch tensors
+    (dev='cpu') so the device-mode orchestration — tensor layouts, gathers,
+    batched fiber blocks, row-block GEMVs — is testable without a GPU. The
+    pair-kernel device functions must be patched to oracle-backed fakes
+    (see patch_device_kernels_with_oracle)."""
+
+    def __init__(self):
+        super().__init__()
+        import torch
+        self.torch = torch
+        self.dev = torch.device("cpu")
+
+    def _t(self, a):
+        import torch
+        if torch.is_tensor(a):
+            return a.to(self.dev)
+        return torch.from_numpy(np.ascontiguousarray(a)).to(self.dev)
+
+
+def patch_device_kernels_with_oracle():
+    """Replace skellysim_amd.evaluator's device kernels with oracle-backed
+    CPU-tensor fakes (identical math by construction). TEST-ONLY; call in a
+    worker subprocess, never in product code."""
+    import torch
+    import oracle
+    import skellysim_amd.evaluator as ev
+
+    def _np(t):
+        return np.ascontiguousarray(t.detach().cpu().numpy())
+
+    ev.stokeslet_device = lambda r, f, t, eta, out=None: torch.from_numpy(
+        oracle.stokeslet(_np(r), _np(f), _np(t), eta))
+    ev.stresslet_device = lambda r, f, t, eta, out=None: torch.from_numpy(
+        oracle.stresslet(_np(r), _np(f), _np(t), eta))
+    ev.oseen_tensor_batched_device = lambda pts, eta=1.0: torch.from_numpy(
+        np.stack([oracle.oseen_tensor(p, eta) for p in _np(pts)]))
+
+

@@ -71,6 +71,46 @@ def test_device_matvec_matches_host_matvec(hip_backend):
     assert rel < 1e-10, rel
 
 
+def test_distributed_device_solve_world1(hip_backend):
+    """DistributedSystemFD's device-resident iteration on the real HIP path
+    (world 1: collectives are identities, everything else — gathered-source
+    matvec, row-block shell GEMVs, device GMRES — is the multi-GPU code)."""
+    import os
+    from skellysim_amd.system_fd import SystemFD, Shell
+    from skellysim_amd.system_dist import DistributedSystemFD
+
+    here = os.path.dirname(os.path.abspath(__file__))
+    fx = np.load(os.path.join(here, "golden", "periphery_sphere_192.npz"))
+    shell = Shell(fx["nodes"], fx["normals"], fx["stresslet_plus_complementary"],
+                  fx["M_inv"])
+    U = np.array([0.05, 0.02, -0.04])
+    bg = lambda r: np.tile(U, (len(r), 1))
+
+    def make_fibers():
+        rng = np.random.default_rng(21)
+        return [straight_fiber(n=32, length=0.5,
+                               direction=rng.uniform(-1, 1, 3),
+                               x0=rng.uniform(-0.25, 0.25, 3),
+                               minus_clamped=(k % 2 == 0))
+                for k in range(5)]
+
+    N = len(fx["nodes"])
+    sys_d = DistributedSystemFD(make_fibers(), eta=1.0, dt=0.05, shell=shell,
+                                shell_rows=(0, N), backend=hip_backend,
+                                background_flow=bg)
+    info = sys_d.solve(tol=1e-11, maxiter=300, device_mode=True)
+    assert info["converged"], info
+
+    sys_s = SystemFD(make_fibers(), eta=1.0, dt=0.05, shell=shell,
+                     backend=hip_backend, background_flow=bg)
+    info2 = sys_s.solve(tol=1e-11, maxiter=300)
+    assert info2["converged"], info2
+
+    rel = np.linalg.norm(sys_d.solution - sys_s.solution) / \
+        np.linalg.norm(sys_s.solution)
+    assert rel < 1e-8, rel
+
+
 def test_hip_matches_oracle_backend_one_solve(hip_backend):
     """One multi-fiber + small-shell solve: HIP backend equals the oracle
     backend to the GMRES tolerance."""
