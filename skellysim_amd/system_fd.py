@@ -228,18 +228,34 @@ class SystemFD:
 
         motor = motor + ext  # total_force_fibers (system.cpp:450)
         v_fib = v_all[:nf_nodes]
+        self._assembled_dev = None
         if self._uniform and self.fibers:
             # batched assembly (operator + RHS + BCs + force operator) —
             # numerically identical to the per-fiber loop (fiber_batch.py)
-            from .fiber_batch import assemble_uniform
+            from .fiber_batch import assemble_uniform, assemble_uniform_t
             n = self.fibers[0].n_nodes
             nf = len(self.fibers)
             flow_b = v_fib.reshape(nf, n, 3).transpose(0, 2, 1)
             motor_b = motor.reshape(nf, n, 3).transpose(0, 2, 1)
             ext_b = ext.reshape(nf, n, 3).transpose(0, 2, 1) \
                 if self.periphery_interaction is not None else None
-            assemble_uniform(self.fibers, dt, eta, flow=flow_b, f_external=motor_b,
-                             bc_force=ext_b)
+            dev = getattr(self.backend, "dev", None)
+            if dev is not None:
+                # assemble on device (the dominant prep cost); materialize
+                # host copies so the host matvec/precond paths stay valid,
+                # and keep the resident tensors for _build_device_operators
+                A_t, RHS_t, F_t = assemble_uniform_t(
+                    self.fibers, dt, eta, flow=flow_b, f_external=motor_b,
+                    bc_force=ext_b, device=dev)
+                A_h, R_h, F_h = (A_t.cpu().numpy(), RHS_t.cpu().numpy(),
+                                 F_t.cpu().numpy())
+                for k, f in enumerate(self.fibers):
+                    f.adopt_operator(A_h[k], R_h[k])
+                    f.force_operator = F_h[k]
+                self._assembled_dev = (A_t, F_t)
+            else:
+                assemble_uniform(self.fibers, dt, eta, flow=flow_b,
+                                 f_external=motor_b, bc_force=ext_b)
         else:
             for f, a, b in self._fiber_node_slices():
                 f.update_RHS(dt, v_fib[a:b].T, motor[a:b].T)
@@ -326,8 +342,11 @@ class SystemFD:
         n = f0.n_nodes
         self._dev = dict(nf=nf, n=n)
         d = self._dev
-        d["A"] = T(np.stack([f.A for f in self.fibers]))
-        d["F"] = T(np.stack([f.force_operator for f in self.fibers]))
+        if getattr(self, "_assembled_dev", None) is not None:
+            d["A"], d["F"] = self._assembled_dev  # already resident from prep
+        else:
+            d["A"] = T(np.stack([f.A for f in self.fibers]))
+            d["F"] = T(np.stack([f.force_operator for f in self.fibers]))
         # self-stokeslets built directly on device (kernels.cpp:146-195)
         from .evaluator import oseen_tensor_batched_device
         pts = T(np.stack([f.x.T for f in self.fibers]))

@@ -115,6 +115,39 @@ def test_batched_assembly_matches_per_fiber_loop():
         assert np.allclose(RHS2[k], Rr, atol=1e-13, rtol=1e-13), k
 
 
+def test_torch_assembly_matches_numpy():
+    """fiber_batch.assemble_uniform_t (the on-device per-timestep assembly)
+    must equal the numpy assembly to fp64 roundoff, across mixed BCs and all
+    optional force inputs."""
+    from skellysim_amd.fiber_batch import assemble_uniform, assemble_uniform_t
+    rng = np.random.default_rng(12)
+    dt, eta, n = 0.05, 1.3, 24
+    fibers = []
+    for k in range(5):
+        f = straight_fiber(n=n, length=0.8 + 0.1 * k,
+                           direction=rng.uniform(-1, 1, 3),
+                           x0=rng.uniform(-1, 1, 3),
+                           minus_clamped=(k % 2 == 0))
+        f.force_scale = -0.05
+        f.update_constants(eta)
+        f.update_derivatives()
+        fibers.append(f)
+    for with_forces in (False, True):
+        flow = np.stack([rng.uniform(-1, 1, (3, n)) for _ in fibers]) \
+            if with_forces else None
+        fe = np.stack([rng.uniform(-1, 1, (3, n)) for _ in fibers]) \
+            if with_forces else None
+        bf = np.stack([rng.uniform(-1, 1, (3, n)) for _ in fibers]) \
+            if with_forces else None
+        A, R, F = assemble_uniform(fibers, dt, eta, flow=flow,
+                                   f_external=fe, bc_force=bf)
+        At, Rt, Ft = assemble_uniform_t(fibers, dt, eta, flow=flow,
+                                        f_external=fe, bc_force=bf)
+        assert np.allclose(A, At.numpy(), atol=1e-13, rtol=1e-13)
+        assert np.allclose(R, Rt.numpy(), atol=1e-12, rtol=1e-13)
+        assert np.allclose(F, Ft.numpy(), atol=1e-13, rtol=1e-13)
+
+
 def test_periphery_repulsion_force():
     """Steric wall force (periphery.cpp:140-162): inward-pointing, decays
     with the gap, zero outside/collided, clamped node skipped."""
