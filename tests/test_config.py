@@ -53,6 +53,61 @@ def test_fibers_inside_committed_periphery_geometry():
         assert np.all(lvl < 1.0)
 
 
+def test_run_sim_cli_end_to_end(tmp_path, monkeypatch):
+    """tools/run_sim.py plumbing on CPU: tiny hand-written reference-format
+    config -> run loop -> reference-format trajectory with the configured
+    dt_write cadence. The CLI's HipBackend is patched to the oracle backend
+    (the CLI itself stays HIP-only)."""
+    import sys
+    import importlib
+    from oracle_backend import OracleBackend
+
+    s = np.linspace(0, 1.0, 16)
+    pts = np.stack([0.1 * np.sin(2 * np.pi * s), np.zeros_like(s), s], axis=1)
+    flat = ", ".join(repr(float(v)) for v in pts.reshape(-1))
+    cfg_path = tmp_path / "tiny.toml"
+    cfg_path.write_text(f"""
+[params]
+eta = 1.0
+dt_initial = 0.1
+dt_write = 0.1
+t_final = 0.25
+gmres_tol = 1e-11
+adaptive_timestep_flag = false
+fiber_type = "FiniteDifference"
+
+[[fibers]]
+length = 1.0
+bending_rigidity = 2.5e-2
+radius = 0.0125
+force_scale = -0.02
+minus_clamped = false
+n_nodes = 16
+x = [{flat}]
+""")
+    out = tmp_path / "traj.out"
+    sys.path.insert(0, os.path.join(os.path.dirname(HERE), "tools"))
+    try:
+        run_sim = importlib.import_module("run_sim")
+    finally:
+        sys.path.pop(0)
+    monkeypatch.setattr(run_sim, "HipBackend", OracleBackend)
+    monkeypatch.setattr(sys, "argv",
+                        ["run_sim.py", "--config-file", str(cfg_path),
+                         "--out", str(out)])
+    run_sim.main()
+
+    from skellysim_amd.listener import Trajectory
+    traj = Trajectory(str(out))
+    assert traj.header["trajversion"] == 1
+    assert len(traj) >= 2                      # t=0.1 and t=0.2 writes
+    times = [f["time"] for f in traj.frames]
+    assert times == sorted(times)
+    f0 = traj.frames[0]["fibers"][1][0]
+    assert np.asarray(f0["x_"]).shape == (16, 3)
+    assert np.isfinite(np.asarray(f0["x_"])).all()
+
+
 def test_periphery_interaction_parsing():
     cfg = load_config(CFG)
     assert periphery_interaction_from(cfg) is None  # flag off in the example
