@@ -10,8 +10,9 @@ src/core/params.cpp:14). Operators are callables (the reference's
 A_fiber_hydro::apply / P_inv_hydro::apply just forward to System::
 apply_matvec / apply_preconditioner, solver_hydro.cpp:23-29,42-48).
 
-The Krylov basis is one (n, restart+1) matrix and each ICGS pass is two
-GEMVs (h = V^T w; w -= V h) — one fused dot-block per pass instead of O(k)
+The Krylov basis is one (restart+1, n) matrix — each basis vector a
+contiguous row, so both ICGS GEMVs (h = V_k w; w -= V_k^T h) take rocBLAS's
+coalesced paths — and each ICGS pass is one fused dot-block instead of O(k)
 scalar ops (scalar torch CPU ops pay a fork-join on many-core hosts).
 
 Distributed: vectors may be rank-local slices of a block-row-distributed
@@ -73,8 +74,13 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
             break
 
         m = min(restart, maxiter - total_iters)
-        V = torch.zeros((n, m + 1), dtype=b.dtype, device=b.device)
-        V[:, 0] = r / beta
+        # Krylov basis stored (m+1, n): each basis vector is a CONTIGUOUS
+        # row, so both ICGS GEMVs (V_k w and V_k^T h) hit rocBLAS's
+        # coalesced paths — the (n, m+1) layout made V_k h a strided
+        # column-slice gemv that dominated whole solves (53% of config-5
+        # GPU time at 278 GB/s; profiles/components_r01.md)
+        V = torch.zeros((m + 1, n), dtype=b.dtype, device=b.device)
+        V[0] = r / beta
         # small dense Hessenberg/Givens state in numpy: scalar torch-CPU ops
         # cost a fork-join on many-core hosts
         H = np.zeros((m + 1, m))
@@ -85,21 +91,21 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
         k_done = 0
 
         for k in range(m):
-            w = matvec(precond(V[:, k].contiguous()))
+            w = matvec(precond(V[k]))
             # ICGS: two classical Gram-Schmidt passes (Belos "ICGS",
             # solver_hydro.cpp:72), each as one fused dot-block + update.
             # All scalars stay on device until ONE combined host transfer
             # per iteration (each .cpu()/float() is a full-stream sync).
-            Vk = V[:, : k + 1]
+            Vk = V[: k + 1]
             hcol_dev = None
             for _ in range(2):
-                h = reduce_(Vk.T @ w)
-                w = w - Vk @ h
+                h = reduce_(Vk @ w)
+                w = w - Vk.T @ h
                 hcol_dev = h if hcol_dev is None else hcol_dev + h
             hk1_dev = torch.sqrt(reduce_(torch.dot(w, w)))
             # device-side normalize without a host read; if hk1 == 0 the
-            # column is never consumed (the loop breaks below)
-            V[:, k + 1] = w / hk1_dev
+            # row is never consumed (the loop breaks below)
+            V[k + 1] = w / hk1_dev
             host_scalars = torch.cat([hcol_dev.reshape(-1),
                                       hk1_dev.reshape(1)]).cpu().numpy()
             H[: k + 1, k] = host_scalars[:-1]
@@ -138,7 +144,7 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
             import scipy.linalg as _scla
             y = _scla.solve_triangular(H[:k_done, :k_done], g[:k_done])
             yt = torch.from_numpy(y).to(dtype=b.dtype, device=b.device)
-            update = V[:, :k_done] @ yt
+            update = V[:k_done].T @ yt
             x = x + precond(update)
         else:
             break
