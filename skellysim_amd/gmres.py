@@ -39,13 +39,23 @@ def _make_reduce(distributed, group=None):
 
 
 def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
-          x0=None, distributed=False, group=None, callback=None):
+          x0=None, distributed=False, group=None, callback=None,
+          sync_cadence=8):
     """Solve A x = b with right-preconditioned GMRES(restart).
 
     matvec(v) -> A v ; precond(v) -> M^-1 v (right preconditioner: solves
     A M^-1 u = b, x = M^-1 u, as the reference does). Convergence: implicit
     residual ||r|| / ||b|| <= tol. Returns (x, info) with info = dict(
     converged, iters, residuals).
+
+    sync_cadence: device->host scalar transfers (each a full-stream sync)
+    are BATCHED every this many iterations — the basis build is fully
+    device-side, so up to sync_cadence Arnoldi steps stay queued on the
+    device stream while the host processes earlier Hessenberg columns.
+    The arithmetic is identical for any cadence (the Givens/residual
+    bookkeeping is merely deferred); at most sync_cadence-1 surplus basis
+    vectors are built past the convergence point and discarded. 1 restores
+    a check per iteration.
     """
     if precond is None:
         precond = lambda v: v
@@ -90,12 +100,14 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
         sn = np.zeros(m)
         k_done = 0
 
+        pending = []  # deferred (device-side) Hessenberg columns
+        stop = False
         for k in range(m):
             w = matvec(precond(V[k]))
             # ICGS: two classical Gram-Schmidt passes (Belos "ICGS",
             # solver_hydro.cpp:72), each as one fused dot-block + update.
-            # All scalars stay on device until ONE combined host transfer
-            # per iteration (each .cpu()/float() is a full-stream sync).
+            # Scalars stay on device; host transfers (full-stream syncs)
+            # are batched every sync_cadence iterations.
             Vk = V[: k + 1]
             hcol_dev = None
             for _ in range(2):
@@ -104,40 +116,55 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
                 hcol_dev = h if hcol_dev is None else hcol_dev + h
             hk1_dev = torch.sqrt(reduce_(torch.dot(w, w)))
             # device-side normalize without a host read; if hk1 == 0 the
-            # row is never consumed (the loop breaks below)
+            # row is never consumed (processing below discards past it)
             V[k + 1] = w / hk1_dev
-            host_scalars = torch.cat([hcol_dev.reshape(-1),
-                                      hk1_dev.reshape(1)]).cpu().numpy()
-            H[: k + 1, k] = host_scalars[:-1]
-            hk1 = host_scalars[-1]
-            H[k + 1, k] = hk1
+            pending.append(torch.cat([hcol_dev.reshape(-1),
+                                      hk1_dev.reshape(1)]))
+            if len(pending) < max(1, sync_cadence) and k != m - 1:
+                continue
 
-            # Givens rotations on the new column (vectorized in numpy)
-            if k:
-                col = H[: k + 1, k]
-                for j in range(k):
-                    t = cs[j] * col[j] + sn[j] * col[j + 1]
-                    col[j + 1] = -sn[j] * col[j] + cs[j] * col[j + 1]
-                    col[j] = t
-            denom = np.sqrt(H[k, k] ** 2 + H[k + 1, k] ** 2)
-            if denom == 0.0:
-                k_done = k
-                break
-            cs[k] = H[k, k] / denom
-            sn[k] = H[k + 1, k] / denom
-            H[k, k] = denom
-            H[k + 1, k] = 0.0
-            g[k + 1] = -sn[k] * g[k]
-            g[k] = cs[k] * g[k]
+            # flush: ONE host transfer, then sequential Givens bookkeeping
+            # for the batched columns
+            scalars = torch.cat(pending).cpu().numpy()
+            k0 = k + 1 - len(pending)
+            off = 0
+            for kk in range(k0, k + 1):
+                col_scalars = scalars[off: off + kk + 2]
+                off += kk + 2
+                H[: kk + 1, kk] = col_scalars[:-1]
+                hk1 = col_scalars[-1]
+                H[kk + 1, kk] = hk1
 
-            total_iters += 1
-            k_done = k + 1
-            resid = float(abs(g[k + 1]) / bnorm)
-            residuals.append(resid)
-            if callback is not None:
-                callback(total_iters, resid)
-            if resid <= tol or hk1 == 0.0:
-                converged = resid <= tol
+                # Givens rotations on the new column
+                if kk:
+                    col = H[: kk + 1, kk]
+                    for j in range(kk):
+                        t = cs[j] * col[j] + sn[j] * col[j + 1]
+                        col[j + 1] = -sn[j] * col[j] + cs[j] * col[j + 1]
+                        col[j] = t
+                denom = np.sqrt(H[kk, kk] ** 2 + H[kk + 1, kk] ** 2)
+                if denom == 0.0 or not np.isfinite(denom):
+                    stop = True
+                    break
+                cs[kk] = H[kk, kk] / denom
+                sn[kk] = H[kk + 1, kk] / denom
+                H[kk, kk] = denom
+                H[kk + 1, kk] = 0.0
+                g[kk + 1] = -sn[kk] * g[kk]
+                g[kk] = cs[kk] * g[kk]
+
+                total_iters += 1
+                k_done = kk + 1
+                resid = float(abs(g[kk + 1]) / bnorm)
+                residuals.append(resid)
+                if callback is not None:
+                    callback(total_iters, resid)
+                if resid <= tol or hk1 == 0.0:
+                    converged = resid <= tol
+                    stop = True
+                    break
+            pending = []
+            if stop:
                 break
 
         if k_done > 0:

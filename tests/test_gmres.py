@@ -41,6 +41,33 @@ def test_right_preconditioned_fewer_iters():
     assert torch.norm(x - x_ref) / torch.norm(x_ref) < 1e-8
 
 
+def test_sync_cadence_equivalence():
+    """Deferred host bookkeeping (sync_cadence > 1) must produce the exact
+    same iterates and residual history as the per-iteration form — the
+    device arithmetic is identical, only the transfer schedule differs.
+    Surplus basis vectors past convergence are discarded."""
+    A, b = make_system(150, 4)
+    x1, i1 = gmres(lambda v: A @ v, b, tol=1e-11, maxiter=300, restart=40,
+                   sync_cadence=1)
+    x8, i8 = gmres(lambda v: A @ v, b, tol=1e-11, maxiter=300, restart=40,
+                   sync_cadence=8)
+    assert i1["converged"] and i8["converged"]
+    assert i1["iters"] == i8["iters"]
+    assert torch.equal(x1, x8)
+    assert np.allclose(i1["residuals"], i8["residuals"], rtol=0, atol=0)
+
+    # also across a restart boundary with a preconditioner
+    A, b = make_system(200, 5, cond="hard")
+    M = torch.linalg.inv(A) + 0.05 * torch.from_numpy(
+        np.random.default_rng(6).uniform(-1, 1, (200, 200)))
+    for cad in (3, 8):
+        xc, ic = gmres(lambda v: A @ v, b, precond=lambda v: M @ v,
+                       tol=1e-10, maxiter=200, restart=30, sync_cadence=cad)
+        x1, i1 = gmres(lambda v: A @ v, b, precond=lambda v: M @ v,
+                       tol=1e-10, maxiter=200, restart=30, sync_cadence=1)
+        assert ic["iters"] == i1["iters"] and torch.equal(xc, x1)
+
+
 def test_restart_path():
     rng = np.random.default_rng(2)
     n = 150
