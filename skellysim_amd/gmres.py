@@ -40,7 +40,7 @@ def _make_reduce(distributed, group=None):
 
 def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
           x0=None, distributed=False, group=None, callback=None,
-          sync_cadence=8):
+          sync_cadence=None):
     """Solve A x = b with right-preconditioned GMRES(restart).
 
     matvec(v) -> A v ; precond(v) -> M^-1 v (right preconditioner: solves
@@ -57,6 +57,9 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
     vectors are built past the convergence point and discarded. 1 restores
     a check per iteration.
     """
+    if sync_cadence is None:
+        import os
+        sync_cadence = int(os.environ.get("SKELLY_GMRES_SYNC_CADENCE", "8"))
     if precond is None:
         precond = lambda v: v
     reduce_ = _make_reduce(distributed, group)
