@@ -41,6 +41,20 @@ def _host_inv(A):
     return torch.from_numpy(scla.inv(A.cpu().numpy())).to(A.device)
 
 
+def _inv_probe_ok(A, X, rtol=1e-6):
+    """Cheap correctness probe for an inverse: A (X v) ~= v for a fixed
+    probe vector (two GEMVs). ROCm 7.2's hipBLAS trsm at ~18000^2 can
+    SILENTLY return corrupt results on some nodes (the same failure that
+    elsewhere raises ALLOC_FAILED or memory-faults) — a wrong M_inv then
+    poisons entire solves, so every inverse is verified before use."""
+    import math
+    n = A.shape[-1]
+    v = torch.linspace(-1.0, 1.0, n, dtype=A.dtype, device=A.device)
+    r = A @ (X @ v) - v
+    rel = float(torch.linalg.vector_norm(r) / torch.linalg.vector_norm(v))
+    return math.isfinite(rel) and rel < rtol
+
+
 def robust_inv(A):
     """torch.linalg.inv hardened against ROCm 7.2 linalg failure modes
     measured on MI355X (all at fp64):
@@ -65,6 +79,8 @@ def robust_inv(A):
             torch.backends.cuda.preferred_linalg_library("default")
         try:
             out = torch.linalg.inv(A)
+            if not _inv_probe_ok(A, out):
+                out = _host_inv(A)
         except RuntimeError:
             out = _host_inv(A)
         finally:
@@ -73,13 +89,23 @@ def robust_inv(A):
         _huge_inv_done = True
         return out
     try:
-        return torch.linalg.inv(A)
+        out = torch.linalg.inv(A)
+        if _inv_probe_ok(A, out):
+            return out
+        import sys
+        print(f"robust_inv: probe failed for {n}x{n} inverse on the default "
+              "backend (silent hipBLAS corruption); retrying via magma",
+              file=sys.stderr, flush=True)
     except RuntimeError as e:
         if "HIPBLAS" not in str(e) or _magma_latched:
             raise
+    if not _magma_latched:
         torch.backends.cuda.preferred_linalg_library("magma")
         _magma_latched = True
-        return torch.linalg.inv(A)
+        out = torch.linalg.inv(A)
+        if _inv_probe_ok(A, out):
+            return out
+    return _host_inv(A)
 
 
 class BatchedLU:
@@ -94,10 +120,39 @@ class BatchedLU:
             raise ValueError(f"expected (batch, m, m), got {tuple(mats.shape)}")
         if mats.dtype != torch.float64:
             raise TypeError("BatchedLU expects fp64")
+        self._host_lus = None
         self.LU, self.pivots = _lu_factor(mats)
+        if mats.is_cuda and not self._probe_ok(mats):
+            # same defensive posture as robust_inv: a silently corrupt
+            # device factorization must not poison solves
+            import sys
+            print("BatchedLU: factor probe failed on the device backend; "
+                  "falling back to host LAPACK factors", file=sys.stderr,
+                  flush=True)
+            import scipy.linalg as scla
+            m_np = mats.cpu().numpy()
+            self._host_lus = [scla.lu_factor(a) for a in m_np]
+            self._dev = mats.device
+
+    def _probe_ok(self, mats, rtol=1e-6):
+        import math
+        m = mats.shape[-1]
+        v = torch.linspace(1.0, 2.0, m, dtype=mats.dtype, device=mats.device)
+        rhs = mats @ v
+        x = self.solve(rhs)
+        rel = float(torch.linalg.vector_norm(x - v) /
+                    torch.linalg.vector_norm(v) / max(1, mats.shape[0]) ** 0.5)
+        return math.isfinite(rel) and rel < rtol
 
     def solve(self, rhs):
         """rhs: (n_fibers, m) or (n_fibers, m, k) -> same shape solution."""
+        if self._host_lus is not None:
+            import numpy as _np
+            import scipy.linalg as scla
+            r = rhs.cpu().numpy()
+            out = _np.stack([scla.lu_solve(lu, b)
+                             for lu, b in zip(self._host_lus, r)])
+            return torch.from_numpy(out).to(self._dev)
         squeeze = rhs.dim() == 2
         if squeeze:
             rhs = rhs.unsqueeze(-1)
