@@ -52,14 +52,21 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
     are BATCHED every this many iterations — the basis build is fully
     device-side, so up to sync_cadence Arnoldi steps stay queued on the
     device stream while the host processes earlier Hessenberg columns.
-    The arithmetic is identical for any cadence (the Givens/residual
-    bookkeeping is merely deferred); at most sync_cadence-1 surplus basis
-    vectors are built past the convergence point and discarded. 1 restores
-    a check per iteration.
+    The host bookkeeping is identical for any cadence (deferred, not
+    changed), and CPU tensors are bitwise-identical across cadences
+    (tests/test_gmres.py). DEFAULT 1: on MI355X, cadence 8 deterministically
+    corrupts large solves (config-5 scale, ~500k unknowns, >70 iterations)
+    in their SECOND solve of a process — implicit residual diverges from the
+    true residual by ~5e-3 — while every pair kernel checks out bitwise in
+    isolation (tools/debug_matvec.py) and the same run is clean at cadence
+    1; root cause (suspected async hazard in a library call that the
+    per-iteration sync masks) not yet isolated. Opt in to >1 via the
+    SKELLY_GMRES_SYNC_CADENCE env var; ~20% faster per iteration when it
+    holds.
     """
     if sync_cadence is None:
         import os
-        sync_cadence = int(os.environ.get("SKELLY_GMRES_SYNC_CADENCE", "8"))
+        sync_cadence = int(os.environ.get("SKELLY_GMRES_SYNC_CADENCE", "1"))
     if precond is None:
         precond = lambda v: v
     reduce_ = _make_reduce(distributed, group)
