@@ -148,6 +148,7 @@ class FiberFD:
         self.penalty_param = float(penalty_param)
         self.beta_tstep = float(beta_tstep)
         self.v_growth = 0.0
+        self.binding_site = (-1, -1)  # (i_body, i_site); body.hpp:31-34
         self.tension = np.zeros(self.n_nodes)
         self.mats = fib_mats(self.n_nodes)
         # free ends by default; update_boundary_conditions may change

@@ -54,6 +54,20 @@ class HipBackend:
         self.torch.cuda.synchronize()
         return G.cpu().numpy()
 
+    def rotlet(self, centers, torques, r_trg, eta):
+        from .evaluator import rotlet_device
+        u = rotlet_device(self._t(centers), self._t(r_trg), self._t(torques), eta)
+        self.torch.cuda.synchronize()
+        return u.cpu().numpy()
+
+    def stresslet_times_normal(self, nodes, normals, eta):
+        """Dense (3n, 3n) operator (kernels.cpp:264-287; eta-independent
+        factor) — the M block of the body preconditioner."""
+        from .evaluator import stresslet_times_normal_device
+        S = stresslet_times_normal_device(self._t(nodes), self._t(normals))
+        self.torch.cuda.synchronize()
+        return S.cpu().numpy()
+
     def batched_lu(self, A_batch):
         """Factor (nf, m, m); returns solve(rhs_batch (nf, m)) -> (nf, m)."""
         from .batched import BatchedLU
@@ -92,11 +106,12 @@ class Shell:
 
 class SystemFD:
     def __init__(self, fibers, eta, dt, shell=None, background_flow=None, backend=None,
-                 periphery_interaction=None):
+                 periphery_interaction=None, bodies=None):
         self.fibers = list(fibers)
         self.eta = float(eta)
         self.dt = float(dt)
         self.shell = shell
+        self.bodies = list(bodies) if bodies is not None else []
         self.background_flow = background_flow  # fn: (n,3) -> (n,3)
         # steric fiber-periphery repulsion (system.cpp:421, params.cpp:18):
         # dict(kind="sphere"|"ellipsoid", f_0=, l_0=, radius=|abc=) or None
@@ -121,15 +136,74 @@ class SystemFD:
     def shell_sol_size(self):
         return 3 * self.shell.n_nodes if self.shell else 0
 
+    @property
+    def body_sol_size(self):
+        return sum(b.solution_size for b in self.bodies)
+
+    @property
+    def body_node_count(self):
+        return sum(b.n_nodes for b in self.bodies)
+
     def fiber_nodes(self):
         return np.concatenate([f.x.T for f in self.fibers], axis=0) \
             if self.fibers else np.zeros((0, 3))
 
+    def body_nodes(self):
+        return np.concatenate([b.nodes for b in self.bodies], axis=0) \
+            if self.bodies else np.zeros((0, 3))
+
+    def body_normals(self):
+        return np.concatenate([b.normals for b in self.bodies], axis=0) \
+            if self.bodies else np.zeros((0, 3))
+
     def all_nodes(self):
+        """Node order [fibers | shell | bodies] (system.cpp:408-418)."""
         parts = [self.fiber_nodes()]
         if self.shell:
             parts.append(self.shell.nodes)
+        if self.bodies:
+            parts.append(self.body_nodes())
         return np.concatenate(parts, axis=0)
+
+    def _body_sol_slices(self):
+        out, off = [], self.fiber_sol_size + self.shell_sol_size
+        for b in self.bodies:
+            out.append((b, off, off + b.solution_size))
+            off += b.solution_size
+        return out
+
+    def _body_node_slices(self):
+        """Slices into the [fibers | shell | bodies] node array."""
+        out, off = [], self.fiber_node_count + \
+            (self.shell.n_nodes if self.shell else 0)
+        for b in self.bodies:
+            out.append((b, off, off + b.n_nodes))
+            off += b.n_nodes
+        return out
+
+    def _body_flow(self, r_trg, x_bodies, forces_torques):
+        """BodyContainer::flow_spherical (body_container.cpp:269-337):
+        double layer of the bodies' surface densities + center Stokeslet of
+        the forces + center rotlet of the torques. x_bodies: the body block
+        of a solution vector (densities per body, point-major, then U/w);
+        forces_torques: (n_bodies, 6) from link conditions (matvec) or the
+        external forces (prep)."""
+        if not self.bodies:
+            return np.zeros_like(r_trg)
+        dens_parts, off = [], 0
+        for b in self.bodies:
+            dens_parts.append(
+                x_bodies[off: off + 3 * b.n_nodes].reshape(b.n_nodes, 3))
+            off += b.solution_size
+        dens = np.concatenate(dens_parts)
+        v = self.backend.stresslet_normal_density(
+            self.body_nodes(), self.body_normals(), dens, r_trg, self.eta)
+        centers = np.stack([b.position for b in self.bodies])
+        v += self.backend.stokeslet(centers, forces_torques[:, 0:3], r_trg,
+                                    self.eta)
+        v += self.backend.rotlet(centers, forces_torques[:, 3:6], r_trg,
+                                 self.eta)
+        return v
 
     def _fiber_slices(self):
         out, off = [], 0
@@ -185,7 +259,7 @@ class SystemFD:
 
     # ---- solver pipeline ------------------------------------------------
     def prep_state_for_solver(self):
-        """system.cpp:396-459 (no bodies, no dynamic instability)."""
+        """system.cpp:396-459 (no dynamic instability)."""
         dt, eta = self.dt, self.eta
         for f in self.fibers:
             f.update_constants(eta)
@@ -225,6 +299,17 @@ class SystemFD:
             v_all += self._fiber_flow(r_all, ext)
         if self.background_flow is not None:
             v_all += self.background_flow(r_all)
+
+        # body caches + external body force/torque flow (system.cpp:427-443)
+        if self.bodies:
+            for b in self.bodies:
+                b.update_cache(eta, self.backend)
+            ext_ft = np.stack([np.concatenate([b.external_force,
+                                               b.external_torque])
+                               for b in self.bodies])
+            if np.any(ext_ft):
+                v_all += self._body_flow(
+                    r_all, np.zeros(self.body_sol_size), ext_ft)
 
         motor = motor + ext  # total_force_fibers (system.cpp:450)
         v_fib = v_all[:nf_nodes]
@@ -274,38 +359,62 @@ class SystemFD:
             self._fiber_lu_solves = solves
 
         rhs_parts = [f.RHS for f in self.fibers]
+        sh_nodes = self.shell.n_nodes if self.shell else 0
         if self.shell:
-            v_shell = v_all[nf_nodes:]
+            v_shell = v_all[nf_nodes: nf_nodes + sh_nodes]
             rhs_parts.append(-v_shell.reshape(-1))  # update_RHS, periphery.cpp:86
+        for b, a, bb in self._body_node_slices():
+            rhs_parts.append(b.update_RHS(v_all[a:bb]))
         self.RHS = np.concatenate(rhs_parts) if rhs_parts else np.zeros(0)
         return self.RHS
 
     def apply_matvec(self, x):
-        """system.cpp:269-324 (fibers + shell)."""
+        """system.cpp:269-324 (fibers + shell + bodies)."""
         nf_nodes = self.fiber_node_count
+        sh_nodes = self.shell.n_nodes if self.shell else 0
         x_fib = x[: self.fiber_sol_size]
-        x_shell = x[self.fiber_sol_size:]
+        x_shell = x[self.fiber_sol_size: self.fiber_sol_size + self.shell_sol_size]
+        x_bodies = x[self.fiber_sol_size + self.shell_sol_size:]
         r_all = self.all_nodes()
 
         fw = self._apply_fiber_force(x_fib)
         v_all = self._fiber_flow(r_all, fw)
 
         if self.shell:
+            # shell double layer flows to fibers AND bodies, not to itself
+            # (system.cpp:302-305,314-316)
             dens = x_shell.reshape(-1, 3)
-            v_shell2fib = self.backend.stresslet_normal_density(
-                self.shell.nodes, self.shell.normals, dens,
-                r_all[:nf_nodes], self.eta) if nf_nodes else np.zeros((0, 3))
-            v_all[:nf_nodes] += v_shell2fib
+            trg = np.concatenate([r_all[:nf_nodes],
+                                  r_all[nf_nodes + sh_nodes:]])
+            if len(trg):
+                v = self.backend.stresslet_normal_density(
+                    self.shell.nodes, self.shell.normals, dens, trg, self.eta)
+                v_all[:nf_nodes] += v[:nf_nodes]
+                v_all[nf_nodes + sh_nodes:] += v[nf_nodes:]
+
+        vel_on_fiber = None
+        if self.bodies:
+            # fiber<->body link conditions + body flow (system.cpp:308-317)
+            from .body import calculate_link_conditions
+            body_vels = np.stack([
+                x[a + 3 * b.n_nodes: bb] for b, a, bb in self._body_sol_slices()])
+            vel_on_fiber, body_ft = calculate_link_conditions(
+                self.fibers, x_fib, body_vels, self.bodies)
+            v_all += self._body_flow(r_all, x_bodies, body_ft)
 
         res = np.zeros_like(x)
         v_fib = v_all[:nf_nodes]
-        for (f, a, b), (_, na, nb) in zip(self._fiber_slices(),
-                                          self._fiber_node_slices()):
-            res[a:b] = f.matvec(x_fib[a:b], v_fib[na:nb].T, None)
+        for i, ((f, a, b), (_, na, nb)) in enumerate(
+                zip(self._fiber_slices(), self._fiber_node_slices())):
+            vb = vel_on_fiber[i] if vel_on_fiber is not None else None
+            res[a:b] = f.matvec(x_fib[a:b], v_fib[na:nb].T, vb)
         if self.shell:
-            v_shell = v_all[nf_nodes:]
-            res[self.fiber_sol_size:] = \
+            v_shell = v_all[nf_nodes: nf_nodes + sh_nodes]
+            res[self.fiber_sol_size: self.fiber_sol_size + self.shell_sol_size] = \
                 self._shell_matvec(x_shell) + v_shell.reshape(-1)
+        for (b, a, bb), (_, na, nb) in zip(self._body_sol_slices(),
+                                           self._body_node_slices()):
+            res[a:bb] = b.matvec(v_all[na:nb], x[a:bb])
         return res
 
     def apply_preconditioner(self, x):
@@ -325,7 +434,11 @@ class SystemFD:
                                             self._fiber_lu_solves):
                     res[a:b] = solve(x_fib[a:b][None])[0]
         if self.shell:
-            res[self.fiber_sol_size:] = self._shell_precond(x[self.fiber_sol_size:])
+            sh = slice(self.fiber_sol_size,
+                       self.fiber_sol_size + self.shell_sol_size)
+            res[sh] = self._shell_precond(x[sh])
+        for b, a, bb in self._body_sol_slices():
+            res[a:bb] = b.apply_preconditioner(x[a:bb])
         return res
 
     # ---- device-resident iteration (uniform fibers + HipBackend) --------
@@ -448,7 +561,7 @@ class SystemFD:
         if restart is None:
             restart = min(200, maxiter)
 
-        device_mode = (self._uniform and self.fibers
+        device_mode = (self._uniform and self.fibers and not self.bodies
                        and isinstance(self.backend, HipBackend))
         if device_mode:
             self._build_device_operators()
@@ -468,10 +581,12 @@ class SystemFD:
         return info
 
     def step(self, tol=1e-10, maxiter=200, restart=None):
-        """system.cpp:482-493 (no bodies): solve then adopt positions."""
+        """system.cpp:482-493: solve then adopt positions."""
         info = self.solve(tol=tol, maxiter=maxiter, restart=restart)
         for f, a, b in self._fiber_slices():
             f.step(self.solution[a:b])
+        for b, a, bb in self._body_sol_slices():
+            b.step(self.dt, self.solution[a:bb])
         return info
 
     # ---- adaptive time-stepping driver (System::run, system.cpp:516-570) --
@@ -513,13 +628,18 @@ class SystemFD:
         return False
 
     def backup(self):
-        """System::backup (system.cpp:495-505): fiber state only here."""
+        """System::backup (system.cpp:495-505)."""
         self._bak = [(f.x.copy(), f.tension.copy()) for f in self.fibers]
+        self._bak_bodies = [(b.position.copy(), b.orientation.copy(),
+                             b.solution_vec.copy()) for b in self.bodies]
 
     def restore(self):
         for f, (x, t) in zip(self.fibers, self._bak):
             f.x = x
             f.tension = t
+        for b, (p, q, s) in zip(self.bodies, self._bak_bodies):
+            b.place(p, q)
+            b.solution_vec = s
 
     def run(self, t_final, adaptive=True, dt_min=1e-4, dt_max=None,
             beta_up=1.2, beta_down=0.5, fiber_error_tol=0.1,

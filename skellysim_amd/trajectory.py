@@ -44,6 +44,19 @@ def eigen(a):
     raise ValueError(f"unsupported shape {a.shape}")
 
 
+def body_map(b):
+    """SphericalBody msgpack map (body_spherical.hpp:77:
+    radius_, position_, orientation_, solution_vec_); quaternions are
+    ['__quat__', w, x, y, z] (eigen_quaternion_plugin.h)."""
+    w, x, y, z = (float(v) for v in b.orientation)
+    return {
+        "radius_": b.radius,
+        "position_": eigen(b.position),
+        "orientation_": ["__quat__", w, x, y, z],
+        "solution_vec_": eigen(np.asarray(b.solution_vec)),
+    }
+
+
 def fiber_map(f):
     """fiber_finite_difference.hpp:160-161 (non-resume field set)."""
     return {
@@ -55,7 +68,7 @@ def fiber_map(f):
         "penalty_param_": f.penalty_param,
         "force_scale_": f.force_scale,
         "beta_tstep_": f.beta_tstep,
-        "binding_site_": [-1, -1],
+        "binding_site_": list(getattr(f, "binding_site", (-1, -1))),
         "tension_": eigen(f.tension),
         "x_": eigen(f.x.T),  # (n, 3) -> 3 x n Eigen
         "minus_clamped_": f.minus_clamped,
@@ -88,7 +101,9 @@ class TrajectoryWriter:
             "rng_state": [],
             "fibers": [FIBERTYPE_FINITEDIFFERENCE,
                        [fiber_map(f) for f in system.fibers]],
-            "bodies": [[], [], []],
+            # [spherical, deformable, ellipsoidal] (body_container.hpp:158)
+            "bodies": [[body_map(b) for b in getattr(system, "bodies", [])],
+                       [], []],
             "shell": {"solution_vec_": eigen(np.asarray(shell_sol))},
         }
         self._fh.write(msgpack.packb(frame))

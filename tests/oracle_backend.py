@@ -34,6 +34,12 @@ class OracleBackend:
 
         return solve
 
+    def rotlet(self, centers, torques, r_trg, eta):
+        return self.oracle.rotlet(centers, r_trg, torques, eta)
+
+    def stresslet_times_normal(self, nodes, normals, eta):
+        return self.oracle.np_stresslet_times_normal(nodes, normals)
+
     def shell_ops(self, A, M_inv):
         return (lambda x: A @ x), (lambda x: M_inv @ x)
 
