@@ -25,11 +25,21 @@ def build_fibers(cfg, eta):
     fibers = []
     for ft in cfg.get("fibers", []):
         x = np.asarray(ft["x"], float).reshape(-1, 3)  # flat col-major 3 x n
-        fibers.append(FiberFD(
+        f = FiberFD(
             x, length=ft["length"], bending_rigidity=ft["bending_rigidity"],
             eta=eta, radius=ft.get("radius", 0.0125),
             force_scale=ft.get("force_scale", 0.0),
-            minus_clamped=bool(ft.get("minus_clamped", False))))
+            minus_clamped=bool(ft.get("minus_clamped", False)))
+        # body attachment (fiber_finite_difference.cpp:44-45); an attached
+        # fiber's minus end is clamped to the body regardless of the flag
+        # (skelly_config.py:277)
+        f.binding_site = (int(ft.get("parent_body", -1)),
+                          int(ft.get("parent_site", -1)))
+        if f.binding_site[0] >= 0:
+            from .fiber_fd import BC_VELOCITY, BC_ANGULAR_VELOCITY
+            f.minus_clamped = True
+            f.bc_minus = (BC_VELOCITY, BC_ANGULAR_VELOCITY)
+        fibers.append(f)
     return fibers
 
 
@@ -49,8 +59,42 @@ def periphery_interaction_from(cfg):
     return None
 
 
+def build_bodies(cfg, body_geometry):
+    """Spherical bodies from reference [[bodies]] tables
+    (skelly_config.py:720-751 / body_spherical.cpp:213-275). body_geometry:
+    an npz path or dict (nodes/normals/weights — the body precompute
+    surface) shared by all bodies, or a list with one entry per body. TOML
+    orientation is Eigen coeffs order [x, y, z, w]
+    (parse_util convert_array<Quaterniond>); nucleation_sites is the flat
+    col-major 3 x m array."""
+    from .body import SphericalBody
+    tables = cfg.get("bodies", [])
+    if not tables:
+        return []
+    geoms = body_geometry if isinstance(body_geometry, (list, tuple)) \
+        else [body_geometry] * len(tables)
+    bodies = []
+    for bt, g in zip(tables, geoms):
+        g = dict(np.load(g)) if isinstance(g, str) else g
+        q = bt.get("orientation", [0.0, 0.0, 0.0, 1.0])
+        sites = np.asarray(bt.get("nucleation_sites", []),
+                           float).reshape(-1, 3)
+        shape = bt.get("shape", "sphere")
+        if shape != "sphere":
+            raise NotImplementedError(f"body shape {shape!r} (round 2)")
+        bodies.append(SphericalBody(
+            g["nodes"], g["normals"],
+            np.asarray(g["weights"]).reshape(-1), bt.get("radius", 1.0),
+            position=bt.get("position", [0.0, 0.0, 0.0]),
+            orientation=(q[3], q[0], q[1], q[2]),
+            nucleation_sites_ref=sites if len(sites) else None,
+            external_force=bt.get("external_force", [0.0, 0.0, 0.0]),
+            external_torque=bt.get("external_torque", [0.0, 0.0, 0.0])))
+    return bodies
+
+
 def build_system(cfg, backend=None, shell_geometry=None, dt=None,
-                 background_flow=None):
+                 background_flow=None, body_geometry=None):
     """SystemFD from a reference config. shell_geometry: npz path or dict
     with nodes/normals/quadrature_weights (the periphery geometry the
     reference's precompute generates; operators are assembled on device)."""
@@ -73,6 +117,9 @@ def build_system(cfg, backend=None, shell_geometry=None, dt=None,
             torch.from_numpy(np.asarray(g["quadrature_weights"])).to(dev))
         shell = Shell(np.asarray(g["nodes"]), np.asarray(g["normals"]), A, M_inv)
 
+    bodies = build_bodies(cfg, body_geometry) if body_geometry is not None \
+        else []
     return SystemFD(fibers, eta=eta, dt=dt, shell=shell,
                     background_flow=background_flow, backend=backend,
-                    periphery_interaction=periphery_interaction_from(cfg))
+                    periphery_interaction=periphery_interaction_from(cfg),
+                    bodies=bodies)

@@ -83,6 +83,7 @@ def fibers_from_frame(frame, eta):
                     penalty_param=fm["penalty_param_"],
                     beta_tstep=fm["beta_tstep_"])
         f.length_prev = fm["length_prev_"]
+        f.binding_site = tuple(fm.get("binding_site_", (-1, -1)))
         t = np.asarray(fm["tension_"], float).reshape(-1)
         f.tension = t if t.size == f.n_nodes else np.zeros(f.n_nodes)
         f.update_derivatives()
@@ -91,10 +92,41 @@ def fibers_from_frame(frame, eta):
     return fibers
 
 
-def velocity_field(frame, targets, eta, compute, shell_geometry=None):
-    """System::velocity_at_targets for fibers (+shell when geometry given).
-    compute: object with stokeslet/stresslet_normal_density (a system_fd
-    backend)."""
+def bodies_from_frame(frame, body_geometry):
+    """Reconstruct spherical bodies from a frame's body maps
+    (body_spherical.hpp:77: radius_, position_, orientation_,
+    solution_vec_) and the reference-frame geometry npz/dict
+    (nodes/normals/weights, as the precompute emits)."""
+    from .body import SphericalBody
+    maps = frame.get("bodies", [[], [], []])[0]
+    bodies = []
+    for m in maps:
+        q = m["orientation_"]
+        assert q[0] == "__quat__"
+        sites = body_geometry.get("nucleation_sites")
+        b = SphericalBody(np.asarray(body_geometry["nodes"], float),
+                          np.asarray(body_geometry["normals"], float),
+                          np.asarray(body_geometry["weights"], float),
+                          float(m["radius_"]),
+                          position=np.asarray(m["position_"], float).reshape(-1),
+                          orientation=np.asarray(q[1:5], float),
+                          nucleation_sites_ref=sites)
+        sol = np.asarray(m["solution_vec_"], float).reshape(-1)
+        if sol.size == b.solution_size:
+            b.solution_vec = sol
+            n3 = 3 * b.n_nodes
+            b.velocity = sol[n3: n3 + 3]
+            b.angular_velocity = sol[n3 + 3: n3 + 6]
+        bodies.append(b)
+    return bodies
+
+
+def velocity_field(frame, targets, eta, compute, shell_geometry=None,
+                   body_geometry=None):
+    """System::velocity_at_targets (system.cpp:330-384) for fibers
+    (+shell/+bodies when geometry is given). compute: object with
+    stokeslet/stresslet_normal_density/rotlet (a system_fd backend)."""
+    targets = np.asarray(targets, float).reshape(-1, 3)
     u = np.zeros((len(targets), 3))
     fibers = fibers_from_frame(frame, eta)
     if fibers:
@@ -116,6 +148,34 @@ def velocity_field(frame, targets, eta, compute, shell_geometry=None):
             u += compute.stresslet_normal_density(
                 shell_geometry["nodes"], shell_geometry["normals"],
                 dens.reshape(-1, 3), targets, eta)
+    bodies = bodies_from_frame(frame, body_geometry) \
+        if body_geometry is not None else []
+    if bodies:
+        from .body import calculate_link_conditions
+        # bc_.flow with the frame solution: double layer + center
+        # stokeslet/rotlet of the LINK forces (system.cpp:349-357; external
+        # forces are zeroed there, per the reference's own comment)
+        nodes = np.concatenate([b.nodes for b in bodies])
+        normals = np.concatenate([b.normals for b in bodies])
+        dens = np.concatenate([b.solution_vec[: 3 * b.n_nodes].reshape(-1, 3)
+                               for b in bodies])
+        u += compute.stresslet_normal_density(nodes, normals, dens, targets, eta)
+        body_vels = np.stack([np.concatenate([b.velocity, b.angular_velocity])
+                              for b in bodies])
+        x_fib = np.concatenate([np.concatenate([f.x.reshape(-1), f.tension])
+                                for f in fibers]) if fibers else np.zeros(0)
+        _, ft = calculate_link_conditions(fibers, x_fib, body_vels, bodies)
+        centers = np.stack([b.position for b in bodies])
+        u += compute.stokeslet(centers, ft[:, 0:3], targets, eta)
+        u += compute.rotlet(centers, ft[:, 3:6], targets, eta)
+        # points inside a body move rigidly (system.cpp:363-371)
+        for b in bodies:
+            dx = targets - b.position[None, :]
+            inside = np.linalg.norm(dx, axis=1) < b.radius
+            if inside.any():
+                u[inside] = b.velocity[None, :] + \
+                    np.cross(np.broadcast_to(b.angular_velocity, (int(inside.sum()), 3)),
+                             dx[inside])
     return u
 
 
@@ -185,7 +245,8 @@ def integrate_streamline(field_fn, x0, dt_init=0.1, t_final=1.0, abs_err=1e-10,
     return {"x": x, "val": val, "time": t.tolist()}
 
 
-def process_streamlines(frame, req, eta, compute, shell_geometry, vortex=False):
+def process_streamlines(frame, req, eta, compute, shell_geometry,
+                        vortex=False, body_geometry=None):
     """process_streamlines / process_vortexlines (listener.cpp:51-74): one
     line per seed column; vortex=True integrates the curl field instead."""
     req = req or {}
@@ -193,7 +254,8 @@ def process_streamlines(frame, req, eta, compute, shell_geometry, vortex=False):
     x0 = np.asarray(x0, float).reshape(-1, 3) if np.size(x0) else np.zeros((0, 3))
     if not len(x0):
         return []
-    field = lambda pts: velocity_field(frame, pts, eta, compute, shell_geometry)
+    field = lambda pts: velocity_field(frame, pts, eta, compute,
+                                       shell_geometry, body_geometry)
     rhs_fn = (lambda pts: vorticity(field, pts)) if vortex else None
     out = []
     for seed in x0:
@@ -210,7 +272,8 @@ def process_streamlines(frame, req, eta, compute, shell_geometry, vortex=False):
     return out
 
 
-def serve(stdin, stdout, traj, compute, eta=1.0, shell_geometry=None):
+def serve(stdin, stdout, traj, compute, eta=1.0, shell_geometry=None,
+          body_geometry=None):
     """The stdin/stdout request loop (listener.cpp:86-137)."""
     while True:
         raw = stdin.read(8)
@@ -235,17 +298,20 @@ def serve(stdin, stdout, traj, compute, eta=1.0, shell_geometry=None):
         vf = cmd.get("velocity_field", {}) or {}
         x = eigen_decode(vf.get("x", []))
         x = np.asarray(x, float).reshape(-1, 3) if np.size(x) else np.zeros((0, 3))
-        u = velocity_field(frame, x, eta, compute, shell_geometry) \
+        u = velocity_field(frame, x, eta, compute, shell_geometry,
+                           body_geometry) \
             if len(x) else np.zeros((0, 3))
         response = {
             "time": float(frame["time"]),
             "i_frame": frame_no,
             "n_frames": len(traj),
             "streamlines": process_streamlines(frame, cmd.get("streamlines"),
-                                               eta, compute, shell_geometry),
+                                               eta, compute, shell_geometry,
+                                               body_geometry=body_geometry),
             "vortexlines": process_streamlines(frame, cmd.get("vortexlines"),
                                                eta, compute, shell_geometry,
-                                               vortex=True),
+                                               vortex=True,
+                                               body_geometry=body_geometry),
             "velocity_field": eigen_encode_3xn(u),
         }
         out = msgpack.packb(response)
@@ -260,6 +326,9 @@ def main():
     ap.add_argument("--eta", type=float, default=1.0)
     ap.add_argument("--shell-geometry", default=None,
                     help="npz with nodes/normals (the precompute geometry)")
+    ap.add_argument("--body-geometry", default=None,
+                    help="npz with nodes/normals/weights (body reference "
+                         "frame geometry from the precompute)")
     args = ap.parse_args()
 
     from .system_fd import HipBackend  # product path: the MI355X engine
@@ -268,9 +337,16 @@ def main():
     if args.shell_geometry:
         fx = np.load(args.shell_geometry)
         shell_geometry = {"nodes": fx["nodes"], "normals": fx["normals"]}
+    body_geometry = None
+    if args.body_geometry:
+        bx = np.load(args.body_geometry)
+        body_geometry = {"nodes": bx["nodes"], "normals": bx["normals"],
+                         "weights": bx["weights"]}
+        if "nucleation_sites" in bx:
+            body_geometry["nucleation_sites"] = bx["nucleation_sites"]
     traj = Trajectory(args.trajectory)
     serve(sys.stdin.buffer, sys.stdout.buffer, traj, compute, eta=args.eta,
-          shell_geometry=shell_geometry)
+          shell_geometry=shell_geometry, body_geometry=body_geometry)
 
 
 if __name__ == "__main__":

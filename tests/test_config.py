@@ -108,6 +108,65 @@ x = [{flat}]
     assert np.isfinite(np.asarray(f0["x_"])).all()
 
 
+def test_body_config_build_and_solve(tmp_path):
+    """[[bodies]] tables (skelly_config.py:720-751) build spherical bodies;
+    a fiber with parent_body/parent_site couples to it; the built system
+    solves on the oracle backend."""
+    import sys
+    sys.path.insert(0, HERE)
+    from oracle_backend import OracleBackend
+    from skellysim_amd.config import load_config, build_bodies, build_system
+
+    fx = np.load(os.path.join(HERE, "golden", "periphery_sphere_192.npz"))
+    R = float(fx["radius"])
+    geom = {"nodes": fx["nodes"], "normals": -fx["normals"],
+            "weights": fx["quadrature_weights"].reshape(-1)}
+    s = np.linspace(0, 1.0, 16)
+    x0 = np.array([R + 0.3, -0.2, 0.1])
+    pts = (x0[None, :] + s[:, None] * np.array([1.0, 0, 0]))
+    flat = ", ".join(repr(float(v)) for v in pts.reshape(-1))
+    site = [R, 0.0, 0.0]
+    cfg_path = tmp_path / "body.toml"
+    cfg_path.write_text(f"""
+[params]
+eta = 1.0
+dt_initial = 0.05
+fiber_type = "FiniteDifference"
+
+[[bodies]]
+shape = "sphere"
+radius = {R!r}
+position = [0.3, -0.2, 0.1]
+orientation = [0.0, 0.0, 0.0, 1.0]
+nucleation_sites = {site!r}
+external_force = [0.0, 0.0, 0.1]
+
+[[fibers]]
+length = 1.0
+bending_rigidity = 2.5e-3
+force_scale = -0.02
+parent_body = 0
+parent_site = 0
+n_nodes = 16
+x = [{flat}]
+""")
+    cfg = load_config(str(cfg_path))
+    bodies = build_bodies(cfg, geom)
+    assert len(bodies) == 1
+    b = bodies[0]
+    assert b.radius == R and np.allclose(b.position, [0.3, -0.2, 0.1])
+    assert np.allclose(b.external_force, [0, 0, 0.1])
+    assert b.nucleation_sites_ref.shape == (1, 3)
+
+    sys_ = build_system(cfg, backend=OracleBackend(), body_geometry=geom)
+    assert len(sys_.bodies) == 1 and len(sys_.fibers) == 1
+    f = sys_.fibers[0]
+    assert f.binding_site == (0, 0) and f.minus_clamped
+    info = sys_.step(tol=1e-11, maxiter=300, restart=150)
+    assert info["converged"], info
+    assert np.linalg.norm(sys_.bodies[0].velocity) > 1e-5  # pushed by F_ext
+
+
 def test_periphery_interaction_parsing():
     cfg = load_config(CFG)
     assert periphery_interaction_from(cfg) is None  # flag off in the example

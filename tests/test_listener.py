@@ -280,6 +280,70 @@ def test_velocity_field_with_shell(tmp_path):
     assert not np.allclose(u2, ref, atol=1e-12)
 
 
+def test_velocity_field_with_bodies(tmp_path):
+    """Body branch of the listener velocity field: the frame's body maps
+    (radius/position/orientation/solution_vec_) + the reference-frame
+    geometry reproduce System::velocity_at_targets' body flow (double layer
+    + link-force center stokeslet/rotlet + interior rigid override)."""
+    import os
+    from skellysim_amd.body import SphericalBody, calculate_link_conditions
+    from skellysim_amd.listener import velocity_field
+    fx = np.load(os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                              "golden", "periphery_sphere_192.npz"))
+    R = float(fx["radius"])
+    geom = {"nodes": fx["nodes"], "normals": -fx["normals"],
+            "weights": fx["quadrature_weights"].reshape(-1),
+            "nucleation_sites": np.array([[R, 0.0, 0.0]])}
+    b = SphericalBody(geom["nodes"], geom["normals"], geom["weights"], R,
+                      position=(0.1, -0.2, 0.3),
+                      nucleation_sites_ref=np.array([[R, 0.0, 0.0]]))
+    s0 = np.linspace(0, 1.0, 16)
+    x = b.nucleation_sites[0][None, :] + s0[:, None] * np.array([1.0, 0, 0])
+    fib = FiberFD(x, length=1.0, bending_rigidity=2.5e-3, eta=1.0,
+                  minus_clamped=True, force_scale=-0.05)
+    fib.binding_site = (0, 0)
+    sys_ = SystemFD([fib], eta=1.0, dt=0.05, bodies=[b],
+                    backend=OracleBackend())
+    path = str(tmp_path / "skelly_sim.out")
+    with TrajectoryWriter(path) as tw:
+        assert sys_.step(tol=1e-11, maxiter=300, restart=150)["converged"]
+        tw.write_frame(sys_, 0.05, 0.05)
+    traj = Trajectory(path)
+    frame = traj.frames[0]
+    # body round trip through the reference wire format
+    bm = frame["bodies"][0][0]
+    assert bm["radius_"] == R and bm["orientation_"][0] == "__quat__"
+    assert np.allclose(np.asarray(bm["position_"]).reshape(-1), b.position)
+    assert np.allclose(np.asarray(bm["solution_vec_"]).reshape(-1),
+                       b.solution_vec)
+
+    targets = np.array([[2.5, 0.4, 0.0], [0.0, 2.2, 1.0],
+                        b.position + [0.3, 0.1, 0.0]])  # last is INSIDE
+    u = velocity_field(frame, targets, 1.0, OracleBackend(),
+                       body_geometry=geom)
+
+    import oracle
+    f = sys_.fibers[0]
+    f.update_derivatives()
+    f.update_force_operator()
+    sol = np.concatenate([f.x.reshape(-1), f.tension])
+    ff = f.force_operator @ sol
+    fn = np.stack([ff[i * f.n_nodes:(i + 1) * f.n_nodes] for i in range(3)],
+                  axis=1)
+    ref = oracle.stokeslet(f.x.T, fn * f.quadrature_weights()[:, None],
+                           targets, 1.0)
+    dens = b.solution_vec[: 3 * b.n_nodes].reshape(-1, 3)
+    f_dl = 2.0 * np.einsum("ni,nj->nij", b.normals, dens).reshape(-1, 9)
+    ref += oracle.stresslet(b.nodes, f_dl, targets, 1.0)
+    bv = np.concatenate([b.velocity, b.angular_velocity])[None]
+    _, ft = calculate_link_conditions([f], sol, bv, [b])
+    ref += oracle.stokeslet(b.position[None], ft[:, 0:3], targets, 1.0)
+    ref += oracle.rotlet(b.position[None], targets, ft[:, 3:6], 1.0)
+    dx = targets[2] - b.position
+    ref[2] = b.velocity + np.cross(b.angular_velocity, dx)
+    assert np.allclose(u, ref, atol=1e-12), np.abs(u - ref).max()
+
+
 def test_invalid_frame_gives_empty_response(tmp_path):
     path, _ = _write_traj(tmp_path)
     reqs = [_request_bytes(99, np.zeros((1, 3))), _request_bytes(0, np.zeros((1, 3)))]

@@ -29,6 +29,9 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--config-file", required=True)
     ap.add_argument("--shell-geometry", default=None)
+    ap.add_argument("--body-geometry", default=None,
+                    help="npz with nodes/normals/weights (the body "
+                         "precompute surface, shared by all bodies)")
     ap.add_argument("--t-final", type=float, default=None,
                     help="override params.t_final")
     ap.add_argument("--out", default="skelly_sim.out")
@@ -42,9 +45,11 @@ def main():
 
     t0 = time.perf_counter()
     sys_ = build_system(cfg, backend=HipBackend(),
-                        shell_geometry=args.shell_geometry)
+                        shell_geometry=args.shell_geometry,
+                        body_geometry=args.body_geometry)
     print(f"system: {len(sys_.fibers)} fibers"
           + (f" + {sys_.shell.n_nodes}-node shell" if sys_.shell else "")
+          + (f" + {len(sys_.bodies)} bodies" if sys_.bodies else "")
           + f", built in {time.perf_counter()-t0:.1f}s", flush=True)
 
     dt_write = p.get("dt_write", 0.1)
