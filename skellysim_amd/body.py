@@ -46,7 +46,11 @@ class SphericalBody:
     nodes_ref/normals_ref: (n, 3) surface geometry in the body frame
     (normals OUTWARD — the opposite of the periphery convention);
     weights: (n,) quadrature weights; nucleation_sites_ref: (m, 3) fiber
-    attachment points in the body frame (body.hpp:31-34)."""
+    attachment points in the body frame (body.hpp:31-34). Attachment sites
+    must sit OFF the quadrature surface (the reference's config radius is
+    the attachment radius, 'the hydrodynamic radius is a bit smaller',
+    skelly_config.py:733-734) — a site coinciding with a quadrature node
+    makes the attached fiber's base stokeslet near-singular."""
 
     def __init__(self, nodes_ref, normals_ref, weights, radius,
                  position=(0.0, 0.0, 0.0), orientation=(1.0, 0.0, 0.0, 0.0),

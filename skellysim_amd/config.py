@@ -43,6 +43,16 @@ def build_fibers(cfg, eta):
     return fibers
 
 
+def periphery_shape_from(cfg):
+    """dict(kind, radius|abc) for collision/binding geometry."""
+    per = cfg.get("periphery", {})
+    if per.get("shape") == "sphere":
+        return dict(kind="sphere", radius=per["radius"])
+    if per.get("shape") == "ellipsoid":
+        return dict(kind="ellipsoid", abc=(per["a"], per["b"], per["c"]))
+    return None
+
+
 def periphery_interaction_from(cfg):
     """fiber-periphery steric params (params.cpp:18,75-78; defaults
     f_0=20, l_0=0.05, params.hpp:46-47) when the flag is on."""
@@ -119,7 +129,14 @@ def build_system(cfg, backend=None, shell_geometry=None, dt=None,
 
     bodies = build_bodies(cfg, body_geometry) if body_geometry is not None \
         else []
+    di = params.get("dynamic_instability")
+    if di is not None and di.get("n_nodes", 0) == 0:
+        di = None
     return SystemFD(fibers, eta=eta, dt=dt, shell=shell,
                     background_flow=background_flow, backend=backend,
                     periphery_interaction=periphery_interaction_from(cfg),
-                    bodies=bodies)
+                    bodies=bodies,
+                    periphery_shape=periphery_shape_from(cfg),
+                    periphery_binding=params.get("periphery_binding"),
+                    dynamic_instability=di,
+                    seed=params.get("seed", 130319))

@@ -128,6 +128,26 @@ BC_VELOCITY = "Velocity"
 BC_ANGULAR_VELOCITY = "AngularVelocity"
 
 
+def points_collide(pc, periphery_shape, threshold=0.0):
+    """Point cloud (3, n) vs periphery collision
+    (SphericalPeriphery::check_collision periphery.cpp:126-133 /
+    EllipsoidalPeriphery periphery.cpp:203-224). periphery_shape: dict with
+    kind="sphere"|"ellipsoid" and radius=/abc=."""
+    if periphery_shape["kind"] == "sphere":
+        r2 = (periphery_shape["radius"] - threshold) ** 2
+        return bool(np.any((pc ** 2).sum(axis=0) >= r2))
+    a, b, c = periphery_shape["abc"]
+    r_s = pc / np.array([a, b, c])[:, None]
+    r_s_mag = np.linalg.norm(r_s, axis=0)
+    phi = np.arctan2(r_s[1], r_s[0] + 1e-12)
+    theta = np.arccos(r_s[2] / (1e-12 + r_s_mag))
+    st = np.sin(theta)
+    rc = np.stack([(a - threshold) * st * np.cos(phi),
+                   (b - threshold) * st * np.sin(phi),
+                   (c - threshold) * np.cos(theta)])
+    return bool(np.any((pc ** 2).sum(axis=0) >= (rc ** 2).sum(axis=0)))
+
+
 class FiberFD:
     """One finite-difference fiber (reference FiberFiniteDifference)."""
 
@@ -381,6 +401,29 @@ class FiberFD:
 
         self.A[4 * np_ - 14:, :] = B
         self.RHS[4 * np_ - 14:] = B_RHS
+
+    def update_boundary_conditions(self, periphery_shape=None,
+                                   periphery_binding=None):
+        """FiberFiniteDifference::update_boundary_conditions
+        (fiber_finite_difference.cpp:74-91): minus end clamped when attached
+        (to a body or by the flag), plus end hinged at the cortex
+        (Velocity, Torque) when periphery binding is active, the plus end's
+        polar angle is inside [polar_angle_start, polar_angle_end], and the
+        fiber collides with the shell within `threshold`."""
+        attached = self.minus_clamped or self.binding_site[0] >= 0
+        self.bc_minus = (BC_VELOCITY, BC_ANGULAR_VELOCITY) if attached \
+            else (BC_FORCE, BC_TORQUE)
+        near = False
+        pb = periphery_binding
+        if pb and pb.get("active", False) and periphery_shape is not None:
+            plus = self.x[:, -1]
+            angle = float(np.arccos(plus[2] / np.linalg.norm(plus)))
+            if pb.get("polar_angle_start", 0.0) <= angle \
+                    <= pb.get("polar_angle_end", 0.5 * np.pi):
+                near = points_collide(self.x, periphery_shape,
+                                      pb.get("threshold", 0.75))
+        self.bc_plus = (BC_VELOCITY, BC_TORQUE) if near \
+            else (BC_FORCE, BC_TORQUE)
 
     def periphery_repulsion(self, kind, f_0=20.0, l_0=0.05, radius=None,
                             abc=None):

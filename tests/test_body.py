@@ -169,7 +169,7 @@ def test_coupled_fiber_body_solve():
     and pushes the body; with the motor off and no external force the system
     stays (numerically) quiescent."""
     nodes, normals, w, R = sphere_fixture()
-    site_ref = np.array([[R, 0.0, 0.0]])
+    site_ref = np.array([[1.1 * R, 0.0, 0.0]])  # off-surface attachment
     n = 16
     for motor, expect_motion in ((-0.05, True), (0.0, False)):
         b = SphericalBody(nodes, normals, w, R,

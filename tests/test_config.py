@@ -122,10 +122,10 @@ def test_body_config_build_and_solve(tmp_path):
     geom = {"nodes": fx["nodes"], "normals": -fx["normals"],
             "weights": fx["quadrature_weights"].reshape(-1)}
     s = np.linspace(0, 1.0, 16)
-    x0 = np.array([R + 0.3, -0.2, 0.1])
+    x0 = np.array([1.1 * R + 0.3, -0.2, 0.1])
     pts = (x0[None, :] + s[:, None] * np.array([1.0, 0, 0]))
     flat = ", ".join(repr(float(v)) for v in pts.reshape(-1))
-    site = [R, 0.0, 0.0]
+    site = [1.1 * R, 0.0, 0.0]  # attachment radius above the surface
     cfg_path = tmp_path / "body.toml"
     cfg_path.write_text(f"""
 [params]

@@ -52,7 +52,7 @@ def test_coupled_fiber_body_hip_matches_oracle(hip_lib_path):
     def build(backend):
         b = SphericalBody(fx["nodes"], -fx["normals"],
                           fx["quadrature_weights"].reshape(-1), R,
-                          nucleation_sites_ref=np.array([[R, 0.0, 0.0]]))
+                          nucleation_sites_ref=np.array([[1.1 * R, 0.0, 0.0]]))
         s0 = np.linspace(0, 1.0, 16)
         x = b.nucleation_sites[0][None, :] + s0[:, None] * np.array([1.0, 0, 0])
         fib = FiberFD(x, length=1.0, bending_rigidity=2.5e-3, eta=1.0,

@@ -1,0 +1,149 @@
+"""CPU tests of dynamic instability (skellysim_amd/instability.py,
+restating src/core/dynamic_instability.cpp) and periphery binding
+(fiber_finite_difference.cpp:74-91)."""
+
+import os
+
+import numpy as np
+import pytest
+
+from skellysim_amd.body import SphericalBody
+from skellysim_amd.fiber_fd import FiberFD, BC_VELOCITY, BC_TORQUE, BC_FORCE
+from skellysim_amd.instability import dynamic_instability
+from skellysim_amd.system_fd import SystemFD
+from oracle_backend import OracleBackend
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+
+
+def straight_fiber(n=8, x0=(2.0, 0, 0), d=(1.0, 0, 0), length=1.0):
+    d = np.asarray(d, float) / np.linalg.norm(d)
+    s = np.linspace(0, length, n)
+    x = np.asarray(x0, float)[None, :] + s[:, None] * d[None, :]
+    return FiberFD(x, length=length, bending_rigidity=2.5e-3, eta=1.0)
+
+
+def make_body(n_sites=8):
+    fx = np.load(os.path.join(HERE, "golden", "periphery_sphere_192.npz"))
+    R = float(fx["radius"])
+    phis = 2 * np.pi * np.arange(n_sites) / n_sites
+    # attachment radius ABOVE the hydrodynamic (quadrature) surface, as the
+    # reference prescribes (skelly_config.py:733-734) — a site coinciding
+    # with a quadrature node makes the fiber-base stokeslet near-singular
+    Ra = 1.1 * R
+    sites = np.stack([Ra * np.cos(phis), Ra * np.sin(phis),
+                      np.zeros(n_sites)], axis=1)
+    return SphericalBody(fx["nodes"], -fx["normals"],
+                         fx["quadrature_weights"].reshape(-1), R,
+                         nucleation_sites_ref=sites)
+
+
+def test_catastrophe_and_growth():
+    """Removal probability 1 - exp(-dt f_cat) (dynamic_instability.cpp:83);
+    survivors grow by dt*v_growth with length_prev updated; plus-pinned
+    fibers get collision-scaled growth and catastrophe (lines 73-78)."""
+    rng = np.random.default_rng(5)
+    params = dict(n_nodes=8, v_growth=0.5, f_catastrophe=2.0,
+                  nucleation_rate=0.0)
+    sys_ = SystemFD([straight_fiber() for _ in range(400)], eta=1.0, dt=0.1,
+                    backend=OracleBackend())
+    stats = dynamic_instability(sys_, params, rng)
+    p_remove = 1 - np.exp(-0.1 * 2.0)
+    assert abs(stats["n_removed"] - 400 * p_remove) < 4 * np.sqrt(400 * p_remove)
+    for f in sys_.fibers:
+        assert f.length == pytest.approx(1.0 + 0.1 * 0.5)
+        assert f.length_prev == 1.0
+        assert f.v_growth == 0.5
+
+    # plus-pinned: v_growth x0.5, f_cat x2 (defaults)
+    rng = np.random.default_rng(6)
+    fibs = [straight_fiber() for _ in range(400)]
+    for f in fibs:
+        f.bc_plus = (BC_VELOCITY, BC_TORQUE)
+    sys2 = SystemFD(fibs, eta=1.0, dt=0.1, backend=OracleBackend())
+    stats2 = dynamic_instability(sys2, params, rng)
+    p2 = 1 - np.exp(-0.1 * 4.0)
+    assert abs(stats2["n_removed"] - 400 * p2) < 4 * np.sqrt(400 * p2)
+    for f in sys2.fibers:
+        assert f.v_growth == 0.25
+        assert f.length == pytest.approx(1.0 + 0.1 * 0.25)
+
+
+def test_nucleation_at_body_sites():
+    """New fibers appear on unoccupied sites only, pointing radially
+    outward at min_length with v_growth 0 (dynamic_instability.cpp:104-193)."""
+    rng = np.random.default_rng(7)
+    b = make_body(n_sites=8)
+    # occupy site 0 with an existing attached fiber
+    f0 = straight_fiber(x0=b.nucleation_sites[0], d=b.nucleation_sites[0])
+    f0.binding_site = (0, 0)
+    params = dict(n_nodes=8, v_growth=0.1, f_catastrophe=0.0,
+                  nucleation_rate=1e6, min_length=0.5, radius=0.02,
+                  bending_rigidity=1e-3)
+    sys_ = SystemFD([f0], eta=1.0, dt=0.1, bodies=[b],
+                    backend=OracleBackend())
+    stats = dynamic_instability(sys_, params, rng)
+    assert stats["n_removed"] == 0
+    assert stats["n_nucleated"] == 7  # all free sites, capped
+    sites_used = sorted(f.binding_site[1] for f in sys_.fibers)
+    assert sites_used == list(range(8))
+    for f in sys_.fibers[1:]:
+        assert f.length == 0.5 and f.v_growth == 0.0 and f.minus_clamped
+        assert f.radius == 0.02 and f.bending_rigidity == 1e-3
+        ib, js = f.binding_site
+        site = b.nucleation_sites[js]
+        assert np.allclose(f.x[:, 0], site, atol=1e-14)
+        u = f.x[:, -1] - f.x[:, 0]
+        out = site - b.position
+        assert np.dot(u, out) > 0  # points outward
+        assert np.allclose(np.cross(u, out), 0, atol=1e-12)
+
+    # disabled when n_nodes == 0
+    assert dynamic_instability(sys_, dict(n_nodes=0), rng) == \
+        dict(n_removed=0, n_nucleated=0)
+
+
+def test_periphery_binding_bcs():
+    """fiber_finite_difference.cpp:74-91: plus end hinges (Velocity, Torque)
+    only when binding is active, the polar angle is in range, and the fiber
+    is within threshold of the shell."""
+    shape = dict(kind="sphere", radius=5.0)
+    pb = dict(active=True, polar_angle_start=0.0,
+              polar_angle_end=0.5 * np.pi, threshold=0.75)
+    # plus end near the shell, angle ~ pi/2 (equator, inside range)
+    f = straight_fiber(n=8, x0=(3.5, 0, 0.1), d=(1.0, 0, 0))
+    f.update_boundary_conditions(shape, pb)
+    assert f.bc_plus == (BC_VELOCITY, BC_TORQUE)
+    assert f.bc_minus[0] == BC_FORCE  # free minus end
+    # same fiber, angle outside the allowed band (pointing down)
+    f2 = straight_fiber(n=8, x0=(0.1, 0, -3.5), d=(0, 0, -1.0))
+    f2.update_boundary_conditions(shape, pb)
+    assert f2.bc_plus == (BC_FORCE, BC_TORQUE)
+    # far from the shell
+    f3 = straight_fiber(n=8, x0=(0.5, 0, 0.5), d=(1.0, 0, 0))
+    f3.update_boundary_conditions(shape, pb)
+    assert f3.bc_plus == (BC_FORCE, BC_TORQUE)
+    # binding off
+    f.update_boundary_conditions(shape, dict(active=False))
+    assert f.bc_plus == (BC_FORCE, BC_TORQUE)
+
+
+@pytest.mark.timeout(600)
+def test_di_system_steps():
+    """End-to-end: a body with sites + dynamic instability; the population
+    changes over steps and every solve converges."""
+    b = make_body(n_sites=6)
+    di = dict(n_nodes=8, v_growth=0.2, f_catastrophe=0.5,
+              nucleation_rate=10.0, min_length=0.4, radius=0.0125,
+              bending_rigidity=2.5e-3)
+    sys_ = SystemFD([], eta=1.0, dt=0.1, bodies=[b], backend=OracleBackend(),
+                    dynamic_instability=di, seed=11)
+    counts = []
+    for _ in range(4):
+        info = sys_.step(tol=1e-10, maxiter=300, restart=150)
+        assert info["converged"], info
+        counts.append(len(sys_.fibers))
+        for f in sys_.fibers:
+            assert np.isfinite(f.x).all()
+    assert counts[-1] > 0  # nucleation happened
+    assert max(counts) <= 6

@@ -293,10 +293,10 @@ def test_velocity_field_with_bodies(tmp_path):
     R = float(fx["radius"])
     geom = {"nodes": fx["nodes"], "normals": -fx["normals"],
             "weights": fx["quadrature_weights"].reshape(-1),
-            "nucleation_sites": np.array([[R, 0.0, 0.0]])}
+            "nucleation_sites": np.array([[1.1 * R, 0.0, 0.0]])}
     b = SphericalBody(geom["nodes"], geom["normals"], geom["weights"], R,
                       position=(0.1, -0.2, 0.3),
-                      nucleation_sites_ref=np.array([[R, 0.0, 0.0]]))
+                      nucleation_sites_ref=np.array([[1.1 * R, 0.0, 0.0]]))
     s0 = np.linspace(0, 1.0, 16)
     x = b.nucleation_sites[0][None, :] + s0[:, None] * np.array([1.0, 0, 0])
     fib = FiberFD(x, length=1.0, bending_rigidity=2.5e-3, eta=1.0,
