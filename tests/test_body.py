@@ -112,6 +112,36 @@ def test_isolated_body_mobility():
     assert np.linalg.norm(w2 - w_ref) / np.linalg.norm(w_ref) < 1e-12
 
 
+@pytest.mark.timeout(600)
+def test_two_body_interaction():
+    """Two equal spheres driven along their line of centers move faster
+    than isolated ones; the speed-up matches the method-of-reflections
+    series U/U0 = 1 + (3/2)(R/d) - (R/d)^3 + O((R/d)^4). This exercises the
+    body-body hydrodynamic coupling through the full solve (measured:
+    1.3472 vs 1.3594 predicted at d=4R; 1.2426 vs 1.2454 at d=6R)."""
+    nodes, normals, w, R = sphere_fixture()
+    eta = 1.0
+    F = np.array([0.0, 0.0, 1.0])
+    U0 = 1.0 / (6 * np.pi * eta * R)
+    for dfac, tol in ((4.0, 0.015), (6.0, 0.005)):
+        d = dfac * R
+        bs = [SphericalBody(nodes, normals, w, R, position=(0, 0, z),
+                            external_force=F) for z in (0.0, d)]
+        s = SystemFD([], eta=eta, dt=0.1, bodies=bs, backend=OracleBackend())
+        info = s.solve(tol=1e-12, maxiter=200)
+        assert info["converged"] and info["iters"] <= 10, info
+        lam = R / d
+        pred = 1 + 1.5 * lam - lam ** 3
+        Uz = []
+        off = 0
+        for b in bs:
+            n3 = 3 * b.n_nodes
+            Uz.append(s.solution[off + n3 + 2])
+            off += b.solution_size
+        assert abs(Uz[0] - Uz[1]) < 1e-4 * U0          # symmetric pair
+        assert abs(Uz[0] / U0 - pred) < tol * pred, (Uz[0] / U0, pred)
+
+
 def test_link_conditions_formulas():
     """body_container.cpp:171-268 restated: check F/L on the body and the
     velocity rows on the fiber against directly-written expressions."""

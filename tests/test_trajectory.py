@@ -69,6 +69,48 @@ def test_structure_matches_serialization_contract(tmp_path):
     assert fr["bodies"] == [[], [], []]
 
 
+def _body_system():
+    from skellysim_amd.body import SphericalBody
+    fx = np.load(os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                              "golden", "periphery_sphere_192.npz"))
+    R = float(fx["radius"])
+    b = SphericalBody(fx["nodes"], -fx["normals"],
+                      fx["quadrature_weights"].reshape(-1), R,
+                      position=(0.2, -0.1, 0.3),
+                      external_force=(0.1, 0.0, -0.2),
+                      external_torque=(0.0, 0.3, 0.0))
+    return SystemFD([], eta=1.0, dt=0.1, bodies=[b],
+                    backend=OracleBackend()), b
+
+
+def test_body_frame_structure(tmp_path):
+    """Spherical bodies serialize per body_spherical.hpp:77 into the first
+    sublist of the [spherical, deformable, ellipsoidal] bodies array."""
+    sys_, b = _body_system()
+    path = str(tmp_path / "skelly_sim.out")
+    with TrajectoryWriter(path) as tw:
+        assert sys_.step(tol=1e-11, maxiter=100)["converged"]
+        tw.write_frame(sys_, 0.1, 0.1)
+    with open(path, "rb") as fh:
+        unp = msgpack.Unpacker(fh, raw=False)
+        next(unp)  # header
+        fr = next(unp)
+    sph, deform, ellip = fr["bodies"]
+    assert deform == [] and ellip == [] and len(sph) == 1
+    bm = sph[0]
+    assert set(bm.keys()) == {"radius_", "position_", "orientation_",
+                              "solution_vec_"}
+    assert bm["radius_"] == b.radius
+    q = bm["orientation_"]
+    assert q[0] == "__quat__" and len(q) == 5
+    assert np.allclose(q[1:], b.orientation)
+    pos = bm["position_"]
+    assert pos[0] == "__eigen__" and np.allclose(pos[3:], b.position)
+    sol = bm["solution_vec_"]
+    assert sol[1] == b.solution_size and sol[2] == 1
+    assert np.allclose(sol[3:], b.solution_vec)
+
+
 @pytest.mark.skipif(not os.path.isdir(REFERENCE), reason="reference tree absent")
 def test_reference_reader_round_trip(tmp_path):
     """The REFERENCE'S OWN reader.py must load our trajectory: times indexed,
@@ -117,3 +159,20 @@ def test_reference_reader_round_trip(tmp_path):
     assert len(fibs) == 2
     assert np.allclose(fibs[0]["x_"], sys_.fibers[0].x.T)
     assert traj["time"] == pytest.approx(times[-1])
+
+    # bodies through the reference reader (flattened sublists,
+    # reader.py:333-341; quaternion decoded as [w, x, y, z] array)
+    os.remove(os.path.join(str(tmp_path), "skelly_sim.out"))
+    sys_b, b = _body_system()
+    with TrajectoryWriter(os.path.join(str(tmp_path), "skelly_sim.out")) as tw:
+        assert sys_b.step(tol=1e-11, maxiter=100)["converged"]
+        tw.write_frame(sys_b, 0.1, 0.1)
+    traj2 = TrajectoryReader(toml_file)
+    traj2.load_frame(0)
+    bodies = traj2["bodies"]
+    assert len(bodies) == 1
+    bm = bodies[0]
+    assert bm["radius_"] == b.radius
+    assert np.allclose(bm["position_"], b.position)
+    assert np.allclose(bm["orientation_"], b.orientation)
+    assert np.allclose(bm["solution_vec_"], b.solution_vec)
