@@ -174,6 +174,25 @@ class SphericalBody:
         return dr2 < (self.radius + other.radius + threshold) ** 2
 
 
+class EllipsoidalBody(SphericalBody):
+    """Rigid ellipsoidal body (src/core/body_ellipsoidal.cpp): the solver
+    machinery is IDENTICAL to the spherical body's (matvec/preconditioner/
+    RHS/step/K are line-for-line the same in the reference; only the
+    config radius becomes the 3-vector axis_length and the collision
+    checks differ — the reference stubs most of those, lines 283-331).
+    The ellipsoidal surface itself enters through the geometry arrays."""
+
+    def __init__(self, nodes_ref, normals_ref, weights, axis_length, **kw):
+        self.axis_length = np.asarray(axis_length, float).reshape(3)
+        super().__init__(nodes_ref, normals_ref, weights,
+                         radius=float(np.max(self.axis_length)), **kw)
+
+    def check_collision(self, other, threshold=0.0):
+        """Not implemented in the reference for ellipsoids
+        (body_ellipsoidal.cpp returns false with a warning); mirrored."""
+        return False
+
+
 def calculate_link_conditions(fibers, x_fib, body_velocities, bodies):
     """Fiber<->body attachment coupling
     (body_container.cpp:171-268): returns

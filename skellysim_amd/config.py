@@ -90,16 +90,24 @@ def build_bodies(cfg, body_geometry):
         sites = np.asarray(bt.get("nucleation_sites", []),
                            float).reshape(-1, 3)
         shape = bt.get("shape", "sphere")
-        if shape != "sphere":
-            raise NotImplementedError(f"body shape {shape!r} (round 2)")
-        bodies.append(SphericalBody(
-            g["nodes"], g["normals"],
-            np.asarray(g["weights"]).reshape(-1), bt.get("radius", 1.0),
-            position=bt.get("position", [0.0, 0.0, 0.0]),
-            orientation=(q[3], q[0], q[1], q[2]),
-            nucleation_sites_ref=sites if len(sites) else None,
-            external_force=bt.get("external_force", [0.0, 0.0, 0.0]),
-            external_torque=bt.get("external_torque", [0.0, 0.0, 0.0])))
+        kw = dict(position=bt.get("position", [0.0, 0.0, 0.0]),
+                  orientation=(q[3], q[0], q[1], q[2]),
+                  nucleation_sites_ref=sites if len(sites) else None,
+                  external_force=bt.get("external_force", [0.0, 0.0, 0.0]),
+                  external_torque=bt.get("external_torque", [0.0, 0.0, 0.0]))
+        nodes = g["nodes"]
+        normals = g["normals"]
+        w = np.asarray(g["weights"]).reshape(-1)
+        if shape == "sphere":
+            bodies.append(SphericalBody(nodes, normals, w,
+                                        bt.get("radius", 1.0), **kw))
+        elif shape == "ellipsoid":
+            from .body import EllipsoidalBody
+            bodies.append(EllipsoidalBody(
+                nodes, normals, w,
+                bt.get("axis_length", [1.0, 1.0, 1.0]), **kw))
+        else:
+            raise NotImplementedError(f"body shape {shape!r}")
     return bodies
 
 

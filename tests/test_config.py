@@ -167,6 +167,25 @@ x = [{flat}]
     assert np.linalg.norm(sys_.bodies[0].velocity) > 1e-5  # pushed by F_ext
 
 
+def test_ellipsoid_body_config():
+    """shape = "ellipsoid" builds an EllipsoidalBody with axis_length
+    (skelly_config.py:735-736)."""
+    from skellysim_amd.config import build_bodies
+    from skellysim_amd.body import EllipsoidalBody
+    g = np.load(os.path.join(HERE, "golden", "ellipsoid_body_nodes.npz"))
+    geom = {"nodes": g["nodes"], "normals": g["normals"],
+            "weights": g["quadrature_weights"].reshape(-1)}
+    cfg = {"bodies": [dict(shape="ellipsoid",
+                           axis_length=[float(g["a"]), float(g["b"]),
+                                        float(g["c"])],
+                           position=[0.5, 0.0, -0.2])]}
+    (b,) = build_bodies(cfg, geom)
+    assert isinstance(b, EllipsoidalBody)
+    assert np.allclose(b.axis_length, [g["a"], g["b"], g["c"]])
+    assert b.radius == float(g["a"])  # max axis as the scalar radius
+    assert not b.check_collision(b)   # reference stubs ellipsoid collisions
+
+
 def test_periphery_interaction_parsing():
     cfg = load_config(CFG)
     assert periphery_interaction_from(cfg) is None  # flag off in the example
