@@ -623,16 +623,33 @@ class SystemFD:
         return err
 
     def check_collision(self, periphery_shape=None, threshold=0.0):
-        """Fiber-periphery collision (f_c_fd.cpp:39-55); minus-clamped
-        fibers skip node 0. periphery_shape: dict like
+        """System::check_collision (system.cpp:576-595): fiber-periphery
+        (f_c_fd.cpp:39-55; minus-clamped fibers skip node 0),
+        body-periphery (spherical shell only: |pos| + R > R_shell -
+        threshold, periphery.cpp:94-97; ellipsoidal shells are stubbed in
+        the reference) and body-body pairs. periphery_shape: dict like
         periphery_interaction."""
         from .fiber_fd import points_collide
-        if periphery_shape is None:
-            return False
-        for f in self.fibers:
-            pc = f.x[:, 1:] if f.minus_clamped else f.x
-            if points_collide(pc, periphery_shape, threshold):
-                return True
+        from .body import EllipsoidalBody
+        if periphery_shape is not None:
+            for f in self.fibers:
+                pc = f.x[:, 1:] if f.minus_clamped else f.x
+                if points_collide(pc, periphery_shape, threshold):
+                    return True
+            if periphery_shape["kind"] == "sphere":
+                for b in self.bodies:
+                    if np.linalg.norm(b.position) + b.radius > \
+                            periphery_shape["radius"] - threshold:
+                        return True
+        for i, b1 in enumerate(self.bodies):
+            for b2 in self.bodies[i + 1:]:
+                # any ellipsoid pairing is stubbed False in the reference
+                # (body_spherical.cpp:328-331, body_ellipsoidal.cpp)
+                if isinstance(b1, EllipsoidalBody) or \
+                        isinstance(b2, EllipsoidalBody):
+                    continue
+                if b1.check_collision(b2, threshold):
+                    return True
         return False
 
     def backup(self):
