@@ -140,11 +140,20 @@ def build_system(cfg, backend=None, shell_geometry=None, dt=None,
     di = params.get("dynamic_instability")
     if di is not None and di.get("n_nodes", 0) == 0:
         di = None
-    return SystemFD(fibers, eta=eta, dt=dt, shell=shell,
-                    background_flow=background_flow, backend=backend,
-                    periphery_interaction=periphery_interaction_from(cfg),
-                    bodies=bodies,
-                    periphery_shape=periphery_shape_from(cfg),
-                    periphery_binding=params.get("periphery_binding"),
-                    dynamic_instability=di,
-                    seed=params.get("seed", 130319))
+    system = SystemFD(fibers, eta=eta, dt=dt, shell=shell,
+                      background_flow=background_flow, backend=backend,
+                      periphery_interaction=periphery_interaction_from(cfg),
+                      bodies=bodies,
+                      periphery_shape=periphery_shape_from(cfg),
+                      periphery_binding=params.get("periphery_binding"),
+                      dynamic_instability=di,
+                      seed=params.get("seed", 130319))
+    # point/background sources (system.cpp:691-695)
+    from .sources import PointSourceContainer, BackgroundSource
+    if cfg.get("point_sources"):
+        system.point_sources = PointSourceContainer.from_config(
+            cfg["point_sources"])
+    if "background" in cfg:
+        system.background_source = BackgroundSource.from_config(
+            cfg["background"])
+    return system

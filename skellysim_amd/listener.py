@@ -122,12 +122,20 @@ def bodies_from_frame(frame, body_geometry):
 
 
 def velocity_field(frame, targets, eta, compute, shell_geometry=None,
-                   body_geometry=None):
+                   body_geometry=None, sources=None):
     """System::velocity_at_targets (system.cpp:330-384) for fibers
-    (+shell/+bodies when geometry is given). compute: object with
+    (+shell/+bodies when geometry is given, +point/background sources when
+    `sources` = (PointSourceContainer|None, BackgroundSource|None) is
+    given; their clock is the frame time). compute: object with
     stokeslet/stresslet_normal_density/rotlet (a system_fd backend)."""
     targets = np.asarray(targets, float).reshape(-1, 3)
     u = np.zeros((len(targets), 3))
+    if sources is not None:
+        psc, bs = sources
+        if psc is not None:
+            u += psc.flow(targets, eta, float(frame.get("time", 0.0)), compute)
+        if bs is not None and bs.is_active():
+            u += bs.flow(targets, eta)
     fibers = fibers_from_frame(frame, eta)
     if fibers:
         r_src, wf = [], []
@@ -246,7 +254,7 @@ def integrate_streamline(field_fn, x0, dt_init=0.1, t_final=1.0, abs_err=1e-10,
 
 
 def process_streamlines(frame, req, eta, compute, shell_geometry,
-                        vortex=False, body_geometry=None):
+                        vortex=False, body_geometry=None, sources=None):
     """process_streamlines / process_vortexlines (listener.cpp:51-74): one
     line per seed column; vortex=True integrates the curl field instead."""
     req = req or {}
@@ -255,7 +263,8 @@ def process_streamlines(frame, req, eta, compute, shell_geometry,
     if not len(x0):
         return []
     field = lambda pts: velocity_field(frame, pts, eta, compute,
-                                       shell_geometry, body_geometry)
+                                       shell_geometry, body_geometry,
+                                       sources)
     rhs_fn = (lambda pts: vorticity(field, pts)) if vortex else None
     out = []
     for seed in x0:
@@ -273,7 +282,7 @@ def process_streamlines(frame, req, eta, compute, shell_geometry,
 
 
 def serve(stdin, stdout, traj, compute, eta=1.0, shell_geometry=None,
-          body_geometry=None):
+          body_geometry=None, sources=None):
     """The stdin/stdout request loop (listener.cpp:86-137)."""
     while True:
         raw = stdin.read(8)
@@ -299,7 +308,7 @@ def serve(stdin, stdout, traj, compute, eta=1.0, shell_geometry=None,
         x = eigen_decode(vf.get("x", []))
         x = np.asarray(x, float).reshape(-1, 3) if np.size(x) else np.zeros((0, 3))
         u = velocity_field(frame, x, eta, compute, shell_geometry,
-                           body_geometry) \
+                           body_geometry, sources) \
             if len(x) else np.zeros((0, 3))
         response = {
             "time": float(frame["time"]),
@@ -307,11 +316,13 @@ def serve(stdin, stdout, traj, compute, eta=1.0, shell_geometry=None,
             "n_frames": len(traj),
             "streamlines": process_streamlines(frame, cmd.get("streamlines"),
                                                eta, compute, shell_geometry,
-                                               body_geometry=body_geometry),
+                                               body_geometry=body_geometry,
+                                               sources=sources),
             "vortexlines": process_streamlines(frame, cmd.get("vortexlines"),
                                                eta, compute, shell_geometry,
                                                vortex=True,
-                                               body_geometry=body_geometry),
+                                               body_geometry=body_geometry,
+                                               sources=sources),
             "velocity_field": eigen_encode_3xn(u),
         }
         out = msgpack.packb(response)
@@ -329,6 +340,9 @@ def main():
     ap.add_argument("--body-geometry", default=None,
                     help="npz with nodes/normals/weights (body reference "
                          "frame geometry from the precompute)")
+    ap.add_argument("--config-file", default=None,
+                    help="reference TOML config; supplies point/background "
+                         "sources and eta for the velocity field")
     args = ap.parse_args()
 
     from .system_fd import HipBackend  # product path: the MI355X engine
@@ -344,9 +358,23 @@ def main():
                          "weights": bx["weights"]}
         if "nucleation_sites" in bx:
             body_geometry["nucleation_sites"] = bx["nucleation_sites"]
+    sources = None
+    eta = args.eta
+    if args.config_file:
+        from .config import load_config
+        from .sources import PointSourceContainer, BackgroundSource
+        cfg = load_config(args.config_file)
+        eta = cfg.get("params", {}).get("eta", eta)
+        psc = PointSourceContainer.from_config(cfg["point_sources"]) \
+            if cfg.get("point_sources") else None
+        bs = BackgroundSource.from_config(cfg["background"]) \
+            if "background" in cfg else None
+        if psc is not None or bs is not None:
+            sources = (psc, bs)
     traj = Trajectory(args.trajectory)
-    serve(sys.stdin.buffer, sys.stdout.buffer, traj, compute, eta=args.eta,
-          shell_geometry=shell_geometry, body_geometry=body_geometry)
+    serve(sys.stdin.buffer, sys.stdout.buffer, traj, compute, eta=eta,
+          shell_geometry=shell_geometry, body_geometry=body_geometry,
+          sources=sources)
 
 
 if __name__ == "__main__":

@@ -60,6 +60,13 @@ class HipBackend:
         self.torch.cuda.synchronize()
         return u.cpu().numpy()
 
+    def oseen_contract(self, r_src, r_trg, density, eta):
+        from .evaluator import oseen_contract_device
+        u = oseen_contract_device(self._t(r_src), self._t(r_trg),
+                                  self._t(density), eta)
+        self.torch.cuda.synchronize()
+        return u.cpu().numpy()
+
     def stresslet_times_normal(self, nodes, normals, eta):
         """Dense (3n, 3n) operator (kernels.cpp:264-287; eta-independent
         factor) — the M block of the body preconditioner."""
@@ -121,6 +128,11 @@ class SystemFD:
         self.periphery_binding = periphery_binding
         self.dynamic_instability = dynamic_instability
         self.rng = np.random.default_rng(seed)  # params.seed default 130319
+        # point/background sources (sources.py); simulation clock for their
+        # lifetimes (properties.time — advanced by run())
+        self.point_sources = None
+        self.background_source = None
+        self.time = 0.0
         self.background_flow = background_flow  # fn: (n,3) -> (n,3)
         # steric fiber-periphery repulsion (system.cpp:421, params.cpp:18):
         # dict(kind="sphere"|"ellipsoid", f_0=, l_0=, radius=|abc=) or None
@@ -321,6 +333,13 @@ class SystemFD:
             v_all += self._fiber_flow(r_all, ext)
         if self.background_flow is not None:
             v_all += self.background_flow(r_all)
+        # point + background sources (psc_.flow / bs_.flow, system.cpp:445-446)
+        if self.point_sources is not None:
+            v_all += self.point_sources.flow(r_all, eta, self.time,
+                                             self.backend)
+        if self.background_source is not None and \
+                self.background_source.is_active():
+            v_all += self.background_source.flow(r_all, eta)
 
         # body caches + external body force/torque flow (system.cpp:427-443)
         if self.bodies:
@@ -682,7 +701,7 @@ class SystemFD:
         accepted step advances the clock by dt_new although the state moved
         by the old dt; here time advances by the dt the step was actually
         taken with."""
-        time_now = 0.0
+        time_now = self.time
         dt_max = dt_max if dt_max is not None else self.dt
         history = []
         while time_now < t_final:
@@ -705,6 +724,7 @@ class SystemFD:
                     raise RuntimeError("Timestep smaller than dt_min")
             if accept:
                 time_now += self.dt
+                self.time = time_now  # properties.time (drives source ttl)
                 history.append(dict(time=time_now, dt=self.dt, iters=info["iters"],
                                     fiber_error=err))
                 if on_accept is not None:

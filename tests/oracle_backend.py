@@ -37,6 +37,9 @@ class OracleBackend:
     def rotlet(self, centers, torques, r_trg, eta):
         return self.oracle.rotlet(centers, r_trg, torques, eta)
 
+    def oseen_contract(self, r_src, r_trg, density, eta):
+        return self.oracle.oseen_contract(r_src, r_trg, density, eta)
+
     def stresslet_times_normal(self, nodes, normals, eta):
         return self.oracle.np_stresslet_times_normal(nodes, normals)
 
