@@ -76,9 +76,15 @@ def fiber_map(f):
 
 
 class TrajectoryWriter:
-    """Streams reference-format frames for a SystemFD run."""
+    """Streams reference-format frames for a SystemFD run. append=True
+    continues an existing trajectory without re-writing the header (the
+    reference's resume path, system.cpp:708-712)."""
 
-    def __init__(self, path, version="skelly-hip-0.1.0", commit="unknown"):
+    def __init__(self, path, version="skelly-hip-0.1.0", commit="unknown",
+                 append=False):
+        if append:
+            self._fh = open(path, "ab")
+            return
         self._fh = open(path, "wb")
         header = {
             "trajversion": TRAJECTORY_VERSION,
@@ -117,3 +123,47 @@ class TrajectoryWriter:
 
     def __exit__(self, *exc):
         self.close()
+
+
+def resume_from_trajectory(system, path):
+    """System::resume_from_trajectory (system.cpp:223-228 +
+    TrajectoryReader::unpack_current_frame): restore the system to the LAST
+    frame of an existing trajectory — fibers rebuilt from their serialized
+    state, body positions/orientations/solutions restored onto the
+    configured bodies, the clock and dt adopted. Returns the frame count.
+    (The reference also restores its Philox rng_state; this engine's numpy
+    Generator stream is a documented deviation, see instability.py.)"""
+    from .listener import Trajectory, fibers_from_frame
+    traj = Trajectory(path)
+    if not traj.frames:
+        raise ValueError(f"no frames in {path}")
+    frame = traj.frames[-1]
+    system.time = float(frame["time"])
+    system.dt = float(frame["dt"])
+    system.fibers = fibers_from_frame(frame, system.eta)
+    system._uniform = all(f.n_nodes == system.fibers[0].n_nodes
+                          for f in system.fibers) if system.fibers else True
+    body_maps = frame.get("bodies", [[], [], []])[0]
+    if len(body_maps) != len(system.bodies):
+        raise ValueError("trajectory body count does not match the system")
+    for b, m in zip(system.bodies, body_maps):
+        q = m["orientation_"]
+        b.place(np.asarray(m["position_"], float).reshape(-1),
+                np.asarray(q[1:5], float))
+        sol = np.asarray(m["solution_vec_"], float).reshape(-1)
+        if sol.size == b.solution_size:
+            b.solution_vec = sol
+            n3 = 3 * b.n_nodes
+            b.velocity = sol[n3: n3 + 3].copy()
+            b.angular_velocity = sol[n3 + 3: n3 + 6].copy()
+    # reassemble the global solution vector (used by frame writes and
+    # velocity fields before the next solve)
+    parts = [np.concatenate([f.x.reshape(-1), f.tension])
+             for f in system.fibers]
+    if system.shell is not None:
+        sh = np.asarray(frame.get("shell", {}).get("solution_vec_",
+                                                   np.zeros(0)), float)
+        parts.append(sh.reshape(-1))
+    parts += [b.solution_vec for b in system.bodies]
+    system.solution = np.concatenate(parts) if parts else np.zeros(0)
+    return len(traj)

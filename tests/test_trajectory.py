@@ -111,6 +111,53 @@ def test_body_frame_structure(tmp_path):
     assert np.allclose(sol[3:], b.solution_vec)
 
 
+@pytest.mark.timeout(600)
+def test_resume_matches_uninterrupted_run(tmp_path):
+    """resume_from_trajectory (system.cpp:223-228): a run interrupted after
+    2 steps and resumed into a FRESH system continues to the same state as
+    an uninterrupted 4-step run (solves are deterministic without dynamic
+    instability), and the appended trajectory has continuous times with a
+    single header."""
+    from skellysim_amd.trajectory import resume_from_trajectory
+    from skellysim_amd.listener import Trajectory
+
+    def fresh():
+        rng = np.random.default_rng(9)
+        s = np.linspace(0, 1.0, 16)
+        x = np.stack([0.1 * np.sin(2 * np.pi * s), np.zeros_like(s), s],
+                     axis=1)
+        fib = FiberFD(x, length=1.0, bending_rigidity=2.5e-2, eta=1.0,
+                      force_scale=-0.02)
+        return SystemFD([fib], eta=1.0, dt=0.1, backend=OracleBackend(),
+                        background_flow=lambda r: np.tile([0.05, 0, 0],
+                                                          (len(r), 1)))
+
+    # uninterrupted 4 steps
+    sA = fresh()
+    sA.run(t_final=0.4, adaptive=False, tol=1e-12)
+
+    # 2 steps, write, resume into a FRESH system, 2 more
+    path = str(tmp_path / "skelly_sim.out")
+    sB = fresh()
+    with TrajectoryWriter(path) as tw:
+        sB.run(t_final=0.2, adaptive=False, tol=1e-12,
+               on_accept=lambda s_, t: tw.write_frame(s_, t, s_.dt))
+    sC = fresh()
+    n = resume_from_trajectory(sC, path)
+    assert n == 2 and sC.time == pytest.approx(0.2)
+    with TrajectoryWriter(path, append=True) as tw:
+        sC.run(t_final=0.4, adaptive=False, tol=1e-12,
+               on_accept=lambda s_, t: tw.write_frame(s_, t, s_.dt))
+
+    assert np.allclose(sC.fibers[0].x, sA.fibers[0].x, atol=1e-12)
+    assert np.allclose(sC.fibers[0].tension, sA.fibers[0].tension, atol=1e-12)
+
+    traj = Trajectory(path)
+    assert traj.header["trajversion"] == 1
+    times = [f["time"] for f in traj.frames]
+    assert times == pytest.approx([0.1, 0.2, 0.3, 0.4])
+
+
 @pytest.mark.skipif(not os.path.isdir(REFERENCE), reason="reference tree absent")
 def test_reference_reader_round_trip(tmp_path):
     """The REFERENCE'S OWN reader.py must load our trajectory: times indexed,

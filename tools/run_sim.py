@@ -36,6 +36,9 @@ def main():
                     help="override params.t_final")
     ap.add_argument("--out", default="skelly_sim.out")
     ap.add_argument("--gmres-tol", type=float, default=None)
+    ap.add_argument("--resume", action="store_true",
+                    help="restore the last frame of --out and append "
+                         "(the reference's skelly_sim --resume)")
     args = ap.parse_args()
 
     cfg = load_config(args.config_file)
@@ -52,10 +55,16 @@ def main():
           + (f" + {len(sys_.bodies)} bodies" if sys_.bodies else "")
           + f", built in {time.perf_counter()-t0:.1f}s", flush=True)
 
-    dt_write = p.get("dt_write", 0.1)
-    next_write = [0.0]
+    if args.resume:
+        from skellysim_amd.trajectory import resume_from_trajectory
+        n_frames = resume_from_trajectory(sys_, args.out)
+        print(f"resumed from frame {n_frames - 1} at t={sys_.time:.4f}",
+              flush=True)
 
-    with TrajectoryWriter(args.out) as tw:
+    dt_write = p.get("dt_write", 0.1)
+    next_write = [sys_.time + dt_write if args.resume else 0.0]
+
+    with TrajectoryWriter(args.out, append=args.resume) as tw:
         def on_accept(s, t):
             if t >= next_write[0]:
                 tw.write_frame(s, t, s.dt)
