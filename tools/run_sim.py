@@ -39,7 +39,16 @@ def main():
     ap.add_argument("--resume", action="store_true",
                     help="restore the last frame of --out and append "
                          "(the reference's skelly_sim --resume)")
+    ap.add_argument("--overwrite", action="store_true",
+                    help="allow clobbering an existing trajectory "
+                         "(skelly_sim.cpp:48-49)")
     args = ap.parse_args()
+    if args.resume and args.overwrite:
+        ap.error("--resume and --overwrite are mutually exclusive")
+    if args.resume and not os.path.exists(args.out):
+        ap.error("--resume supplied without existing trajectory")
+    if not args.resume and not args.overwrite and os.path.exists(args.out):
+        ap.error(f"existing trajectory {args.out!r}; supply --overwrite")
 
     cfg = load_config(args.config_file)
     p = cfg.get("params", {})
