@@ -55,7 +55,10 @@ class SphericalBody:
     def __init__(self, nodes_ref, normals_ref, weights, radius,
                  position=(0.0, 0.0, 0.0), orientation=(1.0, 0.0, 0.0, 0.0),
                  nucleation_sites_ref=None, external_force=(0.0, 0.0, 0.0),
-                 external_torque=(0.0, 0.0, 0.0)):
+                 external_torque=(0.0, 0.0, 0.0),
+                 external_force_type="Linear",
+                 oscillation_amplitude=0.0, oscillation_omega=0.0,
+                 oscillation_phase=0.0):
         self.nodes_ref = np.asarray(nodes_ref, float)
         self.normals_ref = np.asarray(normals_ref, float)
         self.weights = np.asarray(weights, float).reshape(-1)
@@ -66,6 +69,13 @@ class SphericalBody:
                                      else np.zeros((0, 3)))
         self.external_force = np.asarray(external_force, float)
         self.external_torque = np.asarray(external_torque, float)
+        # Linear: constant force; Oscillatory: amplitude*sin(omega*t - phase)
+        # along external_force (body_container.cpp:419-426,
+        # body_spherical.cpp:243-256)
+        self.external_force_type = external_force_type
+        self.oscillation_amplitude = float(oscillation_amplitude)
+        self.oscillation_omega = float(oscillation_omega)
+        self.oscillation_phase = float(oscillation_phase)
         self.velocity = np.zeros(3)
         self.angular_velocity = np.zeros(3)
         self.solution_vec = np.zeros(self.solution_size)
@@ -74,6 +84,15 @@ class SphericalBody:
     @property
     def solution_size(self):
         return 3 * self.n_nodes + 6
+
+    def external_force_at(self, time):
+        """BodyContainer::calculate_external_forces_torques
+        (body_container.cpp:419-426)."""
+        if self.external_force_type == "Oscillatory":
+            return self.oscillation_amplitude * np.sin(
+                self.oscillation_omega * time - self.oscillation_phase) \
+                * self.external_force
+        return self.external_force
 
     def place(self, position, orientation):
         """Move to position/orientation; refresh lab-frame geometry

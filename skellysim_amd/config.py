@@ -94,7 +94,15 @@ def build_bodies(cfg, body_geometry):
                   orientation=(q[3], q[0], q[1], q[2]),
                   nucleation_sites_ref=sites if len(sites) else None,
                   external_force=bt.get("external_force", [0.0, 0.0, 0.0]),
-                  external_torque=bt.get("external_torque", [0.0, 0.0, 0.0]))
+                  external_torque=bt.get("external_torque", [0.0, 0.0, 0.0]),
+                  external_force_type=bt.get("external_force_type", "Linear"),
+                  oscillation_amplitude=bt.get(
+                      "external_oscillation_force_amplitude", 0.0),
+                  # omega = 2 pi * frequency (body_spherical.cpp:252-253)
+                  oscillation_omega=2.0 * np.pi * bt.get(
+                      "external_oscillation_force_frequency", 0.0),
+                  oscillation_phase=bt.get(
+                      "external_oscillation_force_phase", 0.0))
         nodes = g["nodes"]
         normals = g["normals"]
         w = np.asarray(g["weights"]).reshape(-1)

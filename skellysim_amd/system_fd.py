@@ -345,7 +345,7 @@ class SystemFD:
         if self.bodies:
             for b in self.bodies:
                 b.update_cache(eta, self.backend)
-            ext_ft = np.stack([np.concatenate([b.external_force,
+            ext_ft = np.stack([np.concatenate([b.external_force_at(self.time),
                                                b.external_torque])
                                for b in self.bodies])
             if np.any(ext_ft):
