@@ -147,3 +147,27 @@ def test_di_system_steps():
             assert np.isfinite(f.x).all()
     assert counts[-1] > 0  # nucleation happened
     assert max(counts) <= 6
+
+@pytest.mark.timeout(300)
+def test_di_steady_state_occupancy():
+    """Population dynamics (the behavior the reference's
+    dynamic_instability_test.cpp records): each site is a birth-death
+    process — empty -> occupied at ~nucleation_rate*dt, occupied -> empty
+    at 1-exp(-dt*f_cat) — so the steady-state occupied fraction is
+    rate/(rate + f_cat). DI-only loop (no solves), like the reference
+    test's System::dynamic_instability() iterations."""
+    rate, f_cat, dt = 2.0, 2.0, 0.05
+    b = make_body(n_sites=40)
+    params = dict(n_nodes=8, v_growth=0.0, f_catastrophe=f_cat,
+                  nucleation_rate=rate, min_length=0.4)
+    sys_ = SystemFD([], eta=1.0, dt=dt, bodies=[b], backend=OracleBackend())
+    rng = np.random.default_rng(123)
+    counts = []
+    for k in range(800):
+        dynamic_instability(sys_, params, rng)
+        if k >= 200:  # discard the transient
+            counts.append(len(sys_.fibers))
+    occ = np.mean(counts) / 40
+    expected = rate / (rate + f_cat)
+    # ~24000 site-steps with correlation; generous 4-sigma-ish band
+    assert abs(occ - expected) < 0.05, (occ, expected)
