@@ -98,7 +98,10 @@ class TrajectoryWriter:
         self._fh.write(msgpack.packb(header))
 
     def write_frame(self, system, time, dt):
-        shell_sol = (system.solution[system.fiber_sol_size:]
+        # shell block only: [fibers | shell | bodies] layout (system.cpp:78)
+        a = system.fiber_sol_size
+        b = a + system.shell_sol_size
+        shell_sol = (system.solution[a:b]
                      if getattr(system, "solution", None) is not None
                      and system.shell else np.zeros(0))
         frame = {

@@ -223,3 +223,38 @@ def test_reference_reader_round_trip(tmp_path):
     assert np.allclose(bm["position_"], b.position)
     assert np.allclose(bm["orientation_"], b.orientation)
     assert np.allclose(bm["solution_vec_"], b.solution_vec)
+
+
+def test_shell_plus_body_frame_slices(tmp_path):
+    """Regression: with BOTH a shell and bodies, the frame's shell
+    solution_vec_ is exactly the shell block of the [fibers|shell|bodies]
+    solution layout (not the tail including bodies)."""
+    from skellysim_amd.body import SphericalBody
+    from skellysim_amd.system_fd import Shell
+    fx = np.load(os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                              "golden", "periphery_sphere_192.npz"))
+    shell = Shell(fx["nodes"], fx["normals"],
+                  fx["stresslet_plus_complementary"], fx["M_inv"])
+    R = float(fx["radius"])
+    b = SphericalBody(fx["nodes"] * 0.2, -fx["normals"],
+                      fx["quadrature_weights"].reshape(-1) * 0.04, 0.2 * R,
+                      position=(0.2, 0.0, 0.0),
+                      external_force=(0.05, 0.0, 0.0))
+    sys_ = SystemFD([], eta=1.0, dt=0.05, shell=shell, bodies=[b],
+                    backend=OracleBackend())
+    path = str(tmp_path / "skelly_sim.out")
+    with TrajectoryWriter(path) as tw:
+        assert sys_.step(tol=1e-10, maxiter=200, restart=100)["converged"]
+        tw.write_frame(sys_, 0.05, 0.05)
+    with open(path, "rb") as fh:
+        unp = msgpack.Unpacker(fh, raw=False)
+        next(unp)
+        fr = next(unp)
+    sh = fr["shell"]["solution_vec_"]
+    assert sh[1] == sys_.shell_sol_size and sh[2] == 1
+    a = sys_.fiber_sol_size
+    assert np.allclose(sh[3:], sys_.solution[a: a + sys_.shell_sol_size])
+    bm = fr["bodies"][0][0]
+    bsol = bm["solution_vec_"]
+    assert bsol[0] == "__eigen__"
+    assert np.allclose(bsol[3:], sys_.solution[a + sys_.shell_sol_size:])
