@@ -171,3 +171,29 @@ def test_di_steady_state_occupancy():
     expected = rate / (rate + f_cat)
     # ~24000 site-steps with correlation; generous 4-sigma-ish band
     assert abs(occ - expected) < 0.05, (occ, expected)
+
+
+@pytest.mark.timeout(600)
+def test_di_mixed_discretization_solve():
+    """DI-nucleated fibers may have a different n_nodes than pre-existing
+    ones: the NON-uniform per-fiber assembly path plus body link conditions
+    must still solve."""
+    b = make_body(n_sites=5)
+    f0 = FiberFD(b.nucleation_sites[0][None, :]
+                 + np.linspace(0, 0.6, 12)[:, None]
+                 * (b.nucleation_sites[0] / np.linalg.norm(
+                     b.nucleation_sites[0]))[None, :],
+                 length=0.6, bending_rigidity=2.5e-3, eta=1.0,
+                 minus_clamped=True, force_scale=-0.05)
+    f0.binding_site = (0, 0)
+    di = dict(n_nodes=8, v_growth=0.1, f_catastrophe=0.0,
+              nucleation_rate=1e6, min_length=0.4)
+    sys_ = SystemFD([f0], eta=1.0, dt=0.05, bodies=[b],
+                    backend=OracleBackend(), dynamic_instability=di, seed=3)
+    info = sys_.step(tol=1e-10, maxiter=300, restart=150)
+    assert info["converged"], info
+    sizes = sorted({f.n_nodes for f in sys_.fibers})
+    assert sizes == [8, 12] and len(sys_.fibers) == 5
+    assert not sys_._uniform
+    for f in sys_.fibers:
+        assert np.isfinite(f.x).all()
