@@ -11,11 +11,14 @@ Wire contract (listener.cpp:86-137 / reader.py Listener.request):
   size 0 in → terminate; invalid frame -> size-0 response (listener.cpp:110-115)
 
 Velocity-field semantics mirror System::velocity_at_targets
-(system.cpp:330-384) for fibers + shell: fiber forces are
-force_operator @ fiber_solution (apply_fiber_force on the frame's
+(system.cpp:330-384) for fibers + shell + bodies + sources: fiber forces
+are force_operator @ fiber_solution (apply_fiber_force on the frame's
 positions+tension), fiber flow WITHOUT self-subtraction (flow(..., false),
 system.cpp:355), plus the shell double layer from the frame's
-solution_vec_. Streamlines are integrated per the reference's adaptive
+solution_vec_, the bodies' double layer + link-force center
+stokeslet/rotlet with the interior rigid-velocity override
+(system.cpp:349-371), and the configured point/background sources.
+Streamlines are integrated per the reference's adaptive
 5(4) RK scheme (see integrate_streamline); vortex lines are streamlines of
 the central-difference curl of the velocity field (streamline.cpp:16-35,
 115-165), with the singularity stop still on the VELOCITY norm
