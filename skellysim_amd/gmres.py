@@ -92,6 +92,7 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
         residuals.append(float(beta / bnorm))
         if residuals[-1] <= tol:
             converged = True
+            true_resid = residuals[-1]  # this IS the true residual
             break
 
         m = min(restart, maxiter - total_iters)
