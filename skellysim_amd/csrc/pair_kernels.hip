@@ -33,6 +33,9 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdlib>
+#include <map>
+#include <mutex>
+#include <vector>
 
 #define BLOCK 256
 #define TILE 512
@@ -359,6 +362,47 @@ __global__ __launch_bounds__(BLOCK) void reduce_partials(const double *__restric
 
 namespace skelly {
 
+/* Optional persistent split-K workspace (SKELLY_PERSISTENT_WS=1): replaces
+ * the per-launch hipMallocAsync/hipFreeAsync pair with a grow-only buffer
+ * cached per stream. Reuse across launches is safe because launches on one
+ * stream are ordered; outgrown blocks are RETIRED (never freed or reused —
+ * queued work may still reference them; geometric growth bounds the leak at
+ * ~2x the final size). Default OFF: the stream-ordered mempool path is the
+ * round-1-validated behavior, and deep queues of outstanding async
+ * alloc/free pairs are the prime suspect for the sync-cadence corruption
+ * documented in gmres.py / DESIGN.md — flip this on to test that
+ * hypothesis. */
+static bool persistent_ws_enabled() {
+    static const bool on = [] {
+        const char *e = getenv("SKELLY_PERSISTENT_WS");
+        return e && atoi(e) != 0;
+    }();
+    return on;
+}
+
+static double *persistent_ws(hipStream_t stream, size_t bytes) {
+    struct Entry {
+        void *ptr = nullptr;
+        size_t cap = 0;
+    };
+    static std::mutex mu;
+    static std::map<hipStream_t, Entry> cache;
+    static std::vector<void *> retired;
+    std::lock_guard<std::mutex> lk(mu);
+    Entry &e = cache[stream];
+    if (e.cap < bytes) {
+        size_t want = bytes * 2;
+        void *p = nullptr;
+        if (hipMalloc(&p, want) != hipSuccess)
+            return nullptr; /* caller falls back to the async path */
+        if (e.ptr)
+            retired.push_back(e.ptr);
+        e.ptr = p;
+        e.cap = want;
+    }
+    return static_cast<double *>(e.ptr);
+}
+
 static inline int pick_tpt(long long n_trg) {
     /* Prefer 4 targets/thread (amortizes the per-source LDS broadcast reads
      * 4x); occupancy for small target counts is recovered by source
@@ -418,10 +462,17 @@ hipError_t launch_pair(const double *r_src, const double *f_src, const double *r
 
     /* split path: same TPT, gridDim.y source slices into a partial buffer */
     double *workspace = nullptr;
+    bool pooled = false;
     const size_t ws_bytes = (size_t)n_slices * 3 * n_trg * sizeof(double);
-    hipError_t err = hipMallocAsync((void **)&workspace, ws_bytes, stream);
-    if (err != hipSuccess)
-        return err;
+    if (persistent_ws_enabled()) {
+        workspace = persistent_ws(stream, ws_bytes);
+        pooled = workspace != nullptr;
+    }
+    if (!pooled) {
+        hipError_t err = hipMallocAsync((void **)&workspace, ws_bytes, stream);
+        if (err != hipSuccess)
+            return err;
+    }
     dim3 grid((unsigned)blocks, (unsigned)n_slices);
     switch (tpt) {
     case 4:
@@ -440,7 +491,9 @@ hipError_t launch_pair(const double *r_src, const double *f_src, const double *r
     const long long rblocks = (3 * n_trg + BLOCK - 1) / BLOCK;
     hipLaunchKernelGGL((reduce_partials<K>), dim3((unsigned)rblocks), block, 0, stream,
                        workspace, u_trg, n_trg, n_slices, params);
-    err = hipGetLastError();
+    hipError_t err = hipGetLastError();
+    if (pooled)
+        return err;
     hipError_t err2 = hipFreeAsync(workspace, stream);
     return err != hipSuccess ? err : err2;
 }
