@@ -193,3 +193,66 @@ def test_periphery_interaction_parsing():
     pi = periphery_interaction_from(cfg)
     assert pi["kind"] == "ellipsoid" and pi["abc"] == (7.8, 4.16, 4.16)
     assert pi["f_0"] == 20.0 and pi["l_0"] == 0.05
+
+
+def test_run_sim_cli_dynamic_instability(tmp_path, monkeypatch):
+    """End-to-end CLI with a [params.dynamic_instability] table and a body:
+    the adaptive run nucleates/removes fibers mid-run and the trajectory
+    records the changing population."""
+    import sys
+    import importlib
+    sys.path.insert(0, HERE)
+    from oracle_backend import OracleBackend
+
+    fx = np.load(os.path.join(HERE, "golden", "body_sphere_600.npz"))
+    geom_path = tmp_path / "body_geom.npz"
+    np.savez(geom_path, nodes=fx["nodes"], normals=fx["normals"],
+             weights=fx["quadrature_weights"].reshape(-1),
+             nucleation_sites=fx["nucleation_sites"])
+    sites_flat = ", ".join(repr(float(v))
+                           for v in np.asarray(fx["nucleation_sites"])[:6].reshape(-1))
+    cfg_path = tmp_path / "di.toml"
+    cfg_path.write_text(f"""
+[params]
+eta = 1.0
+dt_initial = 0.1
+dt_write = 0.1
+t_final = 0.4
+gmres_tol = 1e-10
+adaptive_timestep_flag = false
+seed = 21
+fiber_type = "FiniteDifference"
+
+[params.dynamic_instability]
+n_nodes = 8
+v_growth = 0.2
+f_catastrophe = 1.0
+nucleation_rate = 20.0
+min_length = 0.4
+
+[[bodies]]
+shape = "sphere"
+radius = {float(fx["radius"])!r}
+nucleation_sites = [{sites_flat}]
+""")
+    out = tmp_path / "traj.out"
+    sys.path.insert(0, os.path.join(os.path.dirname(HERE), "tools"))
+    try:
+        run_sim = importlib.import_module("run_sim")
+    finally:
+        sys.path.pop(0)
+    monkeypatch.setattr(run_sim, "HipBackend", OracleBackend)
+    monkeypatch.setattr(sys, "argv",
+                        ["run_sim.py", "--config-file", str(cfg_path),
+                         "--body-geometry", str(geom_path),
+                         "--out", str(out)])
+    run_sim.main()
+
+    from skellysim_amd.listener import Trajectory
+    traj = Trajectory(str(out))
+    counts = [len(f["fibers"][1]) for f in traj.frames]
+    assert len(traj) >= 3
+    assert max(counts) >= 1          # nucleation happened
+    assert max(counts) <= 6          # bounded by the sites
+    for f in traj.frames:            # bodies recorded every frame
+        assert len(f["bodies"][0]) == 1
