@@ -50,19 +50,72 @@ class DistributedSystemFD(SystemFD):
     """
 
     def __init__(self, fibers, eta, dt, shell=None, shell_rows=None,
-                 background_flow=None, backend=None):
+                 background_flow=None, backend=None, bodies=None):
         # periphery_interaction is deliberately not accepted here: the
         # repulsion-induced flow would need all-gathered fiber sources in
         # prep, which this path does not wire yet (round 2)
         super().__init__(fibers, eta, dt, shell=None,
-                         background_flow=background_flow, backend=backend)
+                         background_flow=background_flow, backend=backend,
+                         bodies=bodies)
         self.shell = shell
         self.shell_rows = shell_rows  # (a, b) node indices owned by this rank
         if shell is not None:
             self._shell_matvec_rows, self._shell_precond_rows = \
                 self.backend.shell_ops(shell.A, shell.M_inv)
 
-    # rank-local solution: [own fibers 4n... | 3*(b-a) shell rows]
+    # Bodies follow the reference's ownership model: the body objects
+    # (geometry) are REPLICATED on every rank — every rank evaluates their
+    # flow — but the body SOLUTION block lives only in rank 0's local
+    # vector and is broadcast per apply (body_container.hpp:99,
+    # system.cpp:309); the link forces each rank computes from its own
+    # fibers are all-reduced (body_container.cpp:131).
+    @staticmethod
+    def _rank():
+        import torch.distributed as dist
+        if dist.is_available() and dist.is_initialized():
+            return dist.get_rank()
+        return 0
+
+    @property
+    def global_body_sol_size(self):
+        return sum(b.solution_size for b in self.bodies)
+
+    @property
+    def body_sol_size(self):
+        return self.global_body_sol_size if self._rank() == 0 else 0
+
+    def _body_sol_slices(self):
+        if self._rank() != 0:
+            return []
+        out, off = [], self.fiber_sol_size + self.shell_sol_size
+        for b in self.bodies:
+            out.append((b, off, off + b.solution_size))
+            off += b.solution_size
+        return out
+
+    def _body_node_slices(self):
+        if self._rank() != 0:
+            return []
+        out, off = [], self.fiber_node_count + self.shell_sol_size // 3
+        for b in self.bodies:
+            out.append((b, off, off + b.n_nodes))
+            off += b.n_nodes
+        return out
+
+    def _bcast_body_solution(self, x):
+        """Rank 0's body block of x -> all ranks (system.cpp:309)."""
+        import torch.distributed as dist
+        full = np.zeros(self.global_body_sol_size)
+        if self._rank() == 0:
+            full[:] = x[self.fiber_sol_size + self.shell_sol_size:]
+        if dist.is_available() and dist.is_initialized() \
+                and dist.get_world_size() > 1:
+            t = torch.from_numpy(full)
+            dist.broadcast(t, src=0)
+            full = t.numpy()
+        return full
+
+    # rank-local solution: [own fibers | own shell rows | bodies if rank 0]
     @property
     def shell_sol_size(self):
         if not self.shell:
@@ -78,29 +131,46 @@ class DistributedSystemFD(SystemFD):
         parts = [self.fiber_nodes()]
         if self.shell:
             parts.append(self._own_shell_nodes())
+        if self.bodies and self._rank() == 0:
+            parts.append(self.body_nodes())
         return np.concatenate(parts, axis=0)
 
     def prep_state_for_solver(self):
         """Identical to the single-rank prep except the shell block is this
-        rank's rows (v_shell slice at its own nodes)."""
+        rank's rows (v_shell slice at its own nodes); the base prep handles
+        the fiber block and (on rank 0) the body rows."""
         saved_shell = self.shell
-        self.shell = None  # run the fiber part of the base prep
-        rhs_fib = super().prep_state_for_solver()
+        self.shell = None  # run the fiber+body part of the base prep
+        rhs_base = super().prep_state_for_solver()
         self.shell = saved_shell
         if self.shell:
+            fib_sz = self.fiber_sol_size
+            rhs_fib, rhs_body = rhs_base[:fib_sz], rhs_base[fib_sz:]
             v_sh = np.zeros((self.shell_sol_size // 3, 3))
             if self.background_flow is not None:
                 v_sh += self.background_flow(self._own_shell_nodes())
-            self.RHS = np.concatenate([rhs_fib, -v_sh.reshape(-1)])
+            if self.bodies:
+                ext_ft = np.stack([
+                    np.concatenate([b.external_force_at(self.time),
+                                    b.external_torque])
+                    for b in self.bodies])
+                if np.any(ext_ft):
+                    v_sh += self._body_flow(
+                        self._own_shell_nodes(),
+                        np.zeros(self.global_body_sol_size), ext_ft)
+            self.RHS = np.concatenate([rhs_fib, -v_sh.reshape(-1), rhs_body])
         else:
-            self.RHS = rhs_fib
+            self.RHS = rhs_base
         return self.RHS
 
     def apply_matvec(self, x):
-        """system.cpp:269-324 with all-gathered sources, local targets."""
+        """system.cpp:269-324 with all-gathered sources, local targets;
+        body solution broadcast from rank 0, link forces all-reduced."""
         nf_nodes = self.fiber_node_count
+        sh_nodes = self.shell_sol_size // 3
         x_fib = x[: self.fiber_sol_size]
-        x_shell_local = x[self.fiber_sol_size:]
+        x_shell_local = x[self.fiber_sol_size:
+                          self.fiber_sol_size + self.shell_sol_size]
         r_local = self.all_nodes()
 
         # fiber sources: gather positions + weighted forces from all ranks
@@ -128,23 +198,53 @@ class DistributedSystemFD(SystemFD):
 
         if self.shell:
             dens_all = _allgather_np(x_shell_local.reshape(-1, 3))
-            if nf_nodes:
-                v_all[:nf_nodes] += self.backend.stresslet_normal_density(
+            # shell double layer flows to fibers AND (rank 0's) body nodes,
+            # not to itself (system.cpp:302-305,314-316)
+            trg_idx = np.r_[np.arange(nf_nodes),
+                            np.arange(nf_nodes + sh_nodes, len(r_local))]
+            if len(trg_idx):
+                v_all[trg_idx] += self.backend.stresslet_normal_density(
                     self.shell.nodes, self.shell.normals, dens_all,
-                    r_local[:nf_nodes], self.eta)
+                    r_local[trg_idx], self.eta)
+
+        vel_on_fiber = None
+        if self.bodies:
+            from .body import calculate_link_conditions
+            import torch.distributed as dist
+            x_bodies_global = self._bcast_body_solution(x)
+            body_vels, off = [], 0
+            for b in self.bodies:
+                body_vels.append(
+                    x_bodies_global[off + 3 * b.n_nodes:
+                                    off + b.solution_size])
+                off += b.solution_size
+            body_vels = np.stack(body_vels)
+            vel_on_fiber, body_ft = calculate_link_conditions(
+                self.fibers, x_fib, body_vels, self.bodies)
+            if dist.is_available() and dist.is_initialized() \
+                    and dist.get_world_size() > 1:
+                t = torch.from_numpy(np.ascontiguousarray(body_ft))
+                dist.all_reduce(t)  # sum link F/T over ranks' fibers
+                body_ft = t.numpy()
+            v_all += self._body_flow(r_local, x_bodies_global, body_ft)
 
         res = np.zeros_like(x)
         v_fib = v_all[:nf_nodes]
-        for (f, a, b), (_, na, nb) in zip(self._fiber_slices(),
-                                          self._fiber_node_slices()):
-            res[a:b] = f.matvec(x_fib[a:b], v_fib[na:nb].T, None)
+        for i, ((f, a, b), (_, na, nb)) in enumerate(
+                zip(self._fiber_slices(), self._fiber_node_slices())):
+            vb = vel_on_fiber[i] if vel_on_fiber is not None else None
+            res[a:b] = f.matvec(x_fib[a:b], v_fib[na:nb].T, vb)
         if self.shell:
             # row-block GEMV against the GLOBAL density (Allgatherv +
             # row-distributed dense ops, periphery.cpp:34-47)
             x_shell_all = dens_all.reshape(-1)
-            v_shell = v_all[nf_nodes:].reshape(-1)
-            res[self.fiber_sol_size:] = \
+            v_shell = v_all[nf_nodes: nf_nodes + sh_nodes].reshape(-1)
+            res[self.fiber_sol_size:
+                self.fiber_sol_size + self.shell_sol_size] = \
                 self._shell_matvec_rows(x_shell_all) + v_shell
+        for (b, a, bb), (_, na, nb) in zip(self._body_sol_slices(),
+                                           self._body_node_slices()):
+            res[a:bb] = b.matvec(v_all[na:nb], x[a:bb])
         return res
 
     def apply_preconditioner(self, x):
@@ -158,9 +258,12 @@ class DistributedSystemFD(SystemFD):
             sol = self._fiber_lu_solve(x_fib.reshape(len(self.fibers), m))
             res[: self.fiber_sol_size] = sol.reshape(-1)
         if self.shell:
-            x_shell_all = _allgather_np(
-                x[self.fiber_sol_size:].reshape(-1, 3)).reshape(-1)
-            res[self.fiber_sol_size:] = self._shell_precond_rows(x_shell_all)
+            sh = slice(self.fiber_sol_size,
+                       self.fiber_sol_size + self.shell_sol_size)
+            x_shell_all = _allgather_np(x[sh].reshape(-1, 3)).reshape(-1)
+            res[sh] = self._shell_precond_rows(x_shell_all)
+        for b, a, bb in self._body_sol_slices():
+            res[a:bb] = b.apply_preconditioner(x[a:bb])
         return res
 
     # ---- device-resident distributed iteration --------------------------
@@ -248,7 +351,10 @@ class DistributedSystemFD(SystemFD):
             and dist.get_world_size() > 1
         if device_mode is None:
             device_mode = bool(self.fibers) and self._uniform \
-                and isinstance(self.backend, HipBackend)
+                and not self.bodies and isinstance(self.backend, HipBackend)
+        if device_mode and self.bodies:
+            raise NotImplementedError(
+                "bodies run on the host matvec path (round 2)")
 
         if device_mode:
             self._build_device_operators()
@@ -268,6 +374,22 @@ class DistributedSystemFD(SystemFD):
         x, info = gmres(mv, b, precond=pc, tol=tol, maxiter=maxiter,
                         restart=restart, distributed=distributed)
         self.solution = x.numpy()
+        return info
+
+    def step(self, tol=1e-10, maxiter=200, restart=None):
+        """Solve then adopt: fibers locally, bodies on EVERY rank from the
+        broadcast rank-0 solution block (bc step Bcast,
+        body_container.cpp:49-60) so the replicated geometry stays
+        consistent."""
+        info = self.solve(tol=tol, maxiter=maxiter, restart=restart)
+        for f, a, b in self._fiber_slices():
+            f.step(self.solution[a:b])
+        if self.bodies:
+            full = self._bcast_body_solution(self.solution)
+            off = 0
+            for b in self.bodies:
+                b.step(self.dt, full[off: off + b.solution_size])
+                off += b.solution_size
         return info
 
 

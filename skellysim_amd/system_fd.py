@@ -349,8 +349,12 @@ class SystemFD:
                                                b.external_torque])
                                for b in self.bodies])
             if np.any(ext_ft):
+                # global-sized zero densities (body_sol_size is rank-local
+                # in the distributed subclass; the flow's sources are the
+                # replicated global bodies)
                 v_all += self._body_flow(
-                    r_all, np.zeros(self.body_sol_size), ext_ft)
+                    r_all, np.zeros(sum(b.solution_size
+                                        for b in self.bodies)), ext_ft)
 
         motor = motor + ext  # total_force_fibers (system.cpp:450)
         v_fib = v_all[:nf_nodes]
