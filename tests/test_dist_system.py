@@ -203,3 +203,78 @@ def test_distributed_body_solve_matches_single_process():
         assert np.allclose(pos, body.position, atol=1e-10)
         assert np.allclose(vel, body.velocity, atol=1e-10)
         assert np.allclose(quat, body.orientation, atol=1e-12)
+
+
+def _dist_triple_worker(rank, world, init_file, q):
+    import torch.distributed as dist
+    from skellysim_amd.system_dist import DistributedSystemFD, distribute_fibers
+    from skellysim_amd.system_fd import Shell
+    from skellysim_amd.sharded import shard_range
+    from oracle_backend import OracleBackend
+
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    try:
+        here = os.path.dirname(os.path.abspath(__file__))
+        fx = np.load(os.path.join(here, "golden", "periphery_sphere_192.npz"))
+        body, fibers = _body_problem()
+        N = len(fx["nodes"])
+        a, b = shard_range(N, world, rank)
+        shell = Shell(fx["nodes"] * 4.0, fx["normals"],
+                      fx["stresslet_plus_complementary"][3 * a: 3 * b],
+                      fx["M_inv"][3 * a: 3 * b])
+        my_fibers = distribute_fibers(fibers, world, rank)
+        sys_ = DistributedSystemFD(my_fibers, eta=1.0, dt=0.05, shell=shell,
+                                   shell_rows=(a, b),
+                                   backend=OracleBackend(), bodies=[body])
+        info = sys_.step(tol=1e-11, maxiter=300, restart=150)
+        q.put((rank, sys_.solution, sys_.fiber_sol_size,
+               sys_.shell_sol_size, info["converged"]))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_distributed_triple_matches_single_process():
+    """Fibers + shell rows + a rank-0 body together (covers the dist
+    matvec's shell-to-body-target path and the [fib|shell|body] local
+    layouts on both ranks)."""
+    with tempfile.TemporaryDirectory() as td:
+        init_file = os.path.join(td, "pg")
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        procs = [ctx.Process(target=_dist_triple_worker,
+                             args=(r, WORLD, init_file, q))
+                 for r in range(WORLD)]
+        for p in procs:
+            p.start()
+        results = {}
+        for _ in range(WORLD):
+            rank, sol, fib_sz, sh_sz, conv = q.get(timeout=250)
+            assert conv
+            results[rank] = (sol, fib_sz, sh_sz)
+        for p in procs:
+            p.join(timeout=60)
+            assert p.exitcode == 0
+
+    from skellysim_amd.system_fd import SystemFD, Shell
+    from oracle_backend import OracleBackend
+    here = os.path.dirname(os.path.abspath(__file__))
+    fx = np.load(os.path.join(here, "golden", "periphery_sphere_192.npz"))
+    body, fibers = _body_problem()
+    shell = Shell(fx["nodes"] * 4.0, fx["normals"],
+                  fx["stresslet_plus_complementary"], fx["M_inv"])
+    sys_ = SystemFD(fibers, eta=1.0, dt=0.05, shell=shell, bodies=[body],
+                    backend=OracleBackend())
+    assert sys_.step(tol=1e-11, maxiter=300, restart=150)["converged"]
+
+    # reassemble [r0_fib | r1_fib | r0_shell | r1_shell | body(from r0)]
+    fib_parts, shell_parts = [], []
+    for r in range(WORLD):
+        sol, fib_sz, sh_sz = results[r]
+        fib_parts.append(sol[:fib_sz])
+        shell_parts.append(sol[fib_sz: fib_sz + sh_sz])
+    body_part = results[0][0][results[0][1] + results[0][2]:]
+    dist_sol = np.concatenate(fib_parts + shell_parts + [body_part])
+    rel = np.linalg.norm(dist_sol - sys_.solution) / np.linalg.norm(sys_.solution)
+    assert rel < 1e-7, rel
