@@ -68,3 +68,39 @@ def test_coupled_fiber_body_hip_matches_oracle(hip_lib_path):
     rel = np.linalg.norm(s_hip.solution - s_cpu.solution) / \
         np.linalg.norm(s_cpu.solution)
     assert rel < 1e-8, rel
+
+
+def test_full_composition_hip_matches_oracle(hip_lib_path):
+    """Shell + body + attached fiber on the HIP backend equals the oracle
+    backend to the GMRES tolerance (the full system.cpp matvec composition
+    with every kernel leg on device)."""
+    from skellysim_amd.body import SphericalBody
+    from skellysim_amd.fiber_fd import FiberFD
+    from skellysim_amd.system_fd import SystemFD, Shell, HipBackend
+    from oracle_backend import OracleBackend
+
+    fx = np.load(os.path.join(HERE, "golden", "periphery_sphere_192.npz"))
+    R = float(fx["radius"])
+
+    def build(backend):
+        shell = Shell(fx["nodes"] * 4.0, fx["normals"],
+                      fx["stresslet_plus_complementary"], fx["M_inv"])
+        b = SphericalBody(fx["nodes"], -fx["normals"],
+                          fx["quadrature_weights"].reshape(-1), R,
+                          nucleation_sites_ref=np.array([[1.1 * R, 0.0, 0.0]]))
+        s0 = np.linspace(0, 1.0, 16)
+        x = b.nucleation_sites[0][None, :] + s0[:, None] * np.array([1.0, 0, 0])
+        fib = FiberFD(x, length=1.0, bending_rigidity=2.5e-3, eta=1.0,
+                      minus_clamped=True, force_scale=-0.05)
+        fib.binding_site = (0, 0)
+        return SystemFD([fib], eta=1.0, dt=0.05, shell=shell, bodies=[b],
+                        backend=backend)
+
+    s_hip = build(HipBackend())
+    info = s_hip.solve(tol=1e-11, maxiter=300, restart=150)
+    assert info["converged"], info
+    s_cpu = build(OracleBackend())
+    assert s_cpu.solve(tol=1e-11, maxiter=300, restart=150)["converged"]
+    rel = np.linalg.norm(s_hip.solution - s_cpu.solution) / \
+        np.linalg.norm(s_cpu.solution)
+    assert rel < 1e-8, rel
