@@ -693,29 +693,30 @@ class SystemFD:
             beta_up=1.2, beta_down=0.5, fiber_error_tol=0.1,
             periphery_shape=None, tol=1e-10, maxiter=300, restart=None,
             on_accept=None):
-        """The reference timestep loop (System::run, system.cpp:516-570):
-        backup -> step -> accept if converged and fiber error within
-        tolerance (grow dt when comfortably inside, params.cpp:9-10 defaults
-        beta_up=1.2, beta_down=0.5), reject+restore+shrink otherwise; abort
-        below dt_min; collision rejects with dt/2. on_accept(system, time)
-        is the trajectory-write hook.
-
-        One deliberate deviation: the reference updates properties.dt to the
-        grown value BEFORE advancing time (system.cpp:553-561), so an
-        accepted step advances the clock by dt_new although the state moved
-        by the old dt; here time advances by the dt the step was actually
-        taken with."""
-        time_now = self.time
+        """The reference timestep loop, replicated EXACTLY
+        (System::run, system.cpp:516-570): backup -> step -> accept if
+        converged and fiber error within tolerance (grow dt when comfortably
+        inside, params.cpp:9-10 defaults beta_up=1.2, beta_down=0.5),
+        reject+restore+shrink otherwise; abort below dt_min; collision
+        rejects with dt/2. Including the reference's time accounting quirk:
+        properties.dt is updated to dt_new BEFORE time advances
+        (system.cpp:554-558), so an accepted step advances the clock by the
+        NEW dt although the state moved by the old one — kept verbatim so
+        the reference's pinned end-to-end tests (e.g. the clamped-buckling
+        peak values) reproduce. on_accept(system, time) is the
+        trajectory-write hook (fires on every accepted step; callers apply
+        the reference's dt_write crossing test, system.cpp:560-561)."""
         dt_max = dt_max if dt_max is not None else self.dt
         history = []
-        while time_now < t_final:
+        while self.time < t_final:
             self.backup()
             info = self.step(tol=tol, maxiter=maxiter, restart=restart)
             err = self.fiber_error()
-            accept = True
             dt_new = self.dt
+            accept = not adaptive
             if adaptive:
                 if info["converged"] and err <= fiber_error_tol:
+                    accept = True
                     if err <= 0.9 * fiber_error_tol:
                         dt_new = min(dt_max, self.dt * beta_up)
                 else:
@@ -726,14 +727,13 @@ class SystemFD:
                     accept = False
                 if dt_new < dt_min:
                     raise RuntimeError("Timestep smaller than dt_min")
+                self.dt = dt_new  # BEFORE the clock advance (system.cpp:554)
             if accept:
-                time_now += self.dt
-                self.time = time_now  # properties.time (drives source ttl)
-                history.append(dict(time=time_now, dt=self.dt, iters=info["iters"],
-                                    fiber_error=err))
+                self.time += self.dt
+                history.append(dict(time=self.time, dt=self.dt,
+                                    iters=info["iters"], fiber_error=err))
                 if on_accept is not None:
-                    on_accept(self, time_now)
+                    on_accept(self, self.time)
             else:
                 self.restore()
-            self.dt = dt_new
         return history

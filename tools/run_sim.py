@@ -71,13 +71,13 @@ def main():
               flush=True)
 
     dt_write = p.get("dt_write", 0.1)
-    next_write = [sys_.time + dt_write if args.resume else 0.0]
 
     with TrajectoryWriter(args.out, append=args.resume) as tw:
         def on_accept(s, t):
-            if t >= next_write[0]:
+            # the reference's write-cadence test (system.cpp:560-561):
+            # write when the clock crosses a dt_write boundary
+            if int(t / dt_write) > int((t - s.dt) / dt_write):
                 tw.write_frame(s, t, s.dt)
-                next_write[0] += dt_write
 
         t0 = time.perf_counter()
         hist = sys_.run(t_final=t_final, adaptive=p.get("adaptive_timestep_flag", True),
