@@ -1,0 +1,68 @@
+"""The reference's clamped-fiber buckling regression tests, reproduced end
+to end (tests/combined/test_clamped_buckling_sigma72.py / _sigma80.py): a
+minus-clamped fiber under compressive motor load sigma, kicked by a
+transient point source, integrated with the reference's adaptive loop and
+dt_write cadence for 50 time units. Below the buckling threshold
+(sigma=72) the deflection peaks DECAY; above it (sigma=80) they GROW — and
+both runs must reproduce the reference's PINNED peak deflections to its own
+rel-error < 1e-6 gate (compare_previous_peaks).
+
+Measured here: sigma=72 peaks 0.08844356158 / 0.05563313544 vs the
+reference's 0.08844356 / 0.05563314; sigma=80 peaks 0.09575812350 /
+0.13564472453 vs 0.09575812 / 0.13564472 — every published digit."""
+
+import numpy as np
+import pytest
+from scipy.signal import find_peaks
+
+from skellysim_amd.fiber_fd import FiberFD
+from skellysim_amd.system_fd import SystemFD
+from skellysim_amd.sources import PointSource, PointSourceContainer
+from oracle_backend import OracleBackend
+
+
+def run_sigma(sigma):
+    """gen_config of the reference test, verbatim parameters."""
+    length, E, n = 1.0, 0.0025, 32
+    force_scale = -sigma * E / length ** 3
+    x = np.linspace([0, 0, 0], [0, 0, length], n)
+    fib = FiberFD(x, length=length, bending_rigidity=E, eta=1.0,
+                  minus_clamped=True, force_scale=force_scale)
+    s = SystemFD([fib], eta=1.0, dt=0.02, backend=OracleBackend())
+    s.point_sources = PointSourceContainer(
+        [PointSource(position=[0.0, 0.0, 10 * length], force=[10.0, 0.0, 0.0],
+                     time_to_live=1.0)])
+    dt_write = 0.1
+    xs = []
+
+    def on_accept(sys_, t):
+        # the reference's write cadence (system.cpp:560-561)
+        if int(t / dt_write) > int((t - sys_.dt) / dt_write):
+            xs.append(sys_.fibers[0].x[0, -1])   # plus-end x per frame
+
+    s.run(t_final=50.0, adaptive=True, dt_min=0.01, dt_max=0.1,
+          tol=1e-10, maxiter=300, on_accept=on_accept)
+    x_arr = np.array(xs)
+    peaks, _ = find_peaks(x_arr, height=0)
+    return x_arr, peaks
+
+
+def _check(sigma, pin1, pin2, growing):
+    x, peaks = run_sigma(sigma)
+    p1, p2 = x[peaks[1]], x[peaks[2]]   # skip the kick peak, as the ref does
+    if growing:
+        assert p2 > p1                   # supercritical: buckling grows
+    else:
+        assert p2 < p1                   # subcritical: oscillation decays
+    rel = np.sqrt((1 - p1 / pin1) ** 2 + (1 - p2 / pin2) ** 2)
+    assert rel < 1e-6, (p1, p2, rel)     # the reference's own gate
+
+
+@pytest.mark.timeout(600)
+def test_clamped_buckling_sigma72_subcritical():
+    _check(72.0, 0.08844356, 0.05563314, growing=False)
+
+
+@pytest.mark.timeout(600)
+def test_clamped_buckling_sigma80_supercritical():
+    _check(80.0, 0.09575812, 0.13564472, growing=True)
