@@ -25,6 +25,7 @@ the central-difference curl of the velocity field (streamline.cpp:16-35,
 (streamline.cpp:51, same observer for both line kinds)."""
 
 import argparse
+import os
 import struct
 import sys
 
@@ -346,7 +347,15 @@ def main():
     ap.add_argument("--config-file", default=None,
                     help="reference TOML config; supplies point/background "
                          "sources and eta for the velocity field")
+    ap.add_argument("--listen", action="store_true",
+                    help="compatibility no-op: the reference client spawns "
+                         "`<binary> --listen` (reader.py:155); this CLI "
+                         "always listens")
     args = ap.parse_args()
+    # the reference convention: skelly_config.toml in the working directory
+    # (reader.py:138); pick it up when present and not overridden
+    if args.config_file is None and os.path.exists("skelly_config.toml"):
+        args.config_file = "skelly_config.toml"
 
     from .system_fd import HipBackend  # product path: the MI355X engine
     compute = HipBackend()

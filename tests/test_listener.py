@@ -351,3 +351,44 @@ def test_invalid_frame_gives_empty_response(tmp_path):
     assert res[0] is None          # size-0 response (listener.cpp:110-115)
     assert res[1] is not None      # listener continues serving
     assert res[1]["i_frame"] == 0
+
+
+@pytest.mark.timeout(300)
+def test_listener_cli_subprocess(tmp_path):
+    """The CLI process path the reference's Listener client drives
+    (reader.py:155 spawns `<binary> --listen`): tools/skelly_sim_hip serves
+    the wire protocol over real pipes. Metadata-only request (no compute,
+    so it runs without a GPU)."""
+    import os
+    import subprocess
+    path, sys_ = _write_traj(tmp_path)
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    launcher = os.path.join(repo, "tools", "skelly_sim_hip")
+    cmd = {
+        "frame_no": 1, "evaluator": "GPU",
+        "streamlines": {"dt_init": 0.1, "t_final": 1.0, "abs_err": 1e-10,
+                        "rel_err": 1e-6, "back_integrate": True,
+                        "x0": np.zeros((0, 3))},
+        "vortexlines": {"dt_init": 0.1, "t_final": 1.0, "abs_err": 1e-10,
+                        "rel_err": 1e-6, "back_integrate": True,
+                        "x0": np.zeros((0, 3))},
+        "velocity_field": {"x": np.zeros((0, 3))},
+    }
+    msg = msgpack.packb(cmd, default=_ndencode)
+    proc = subprocess.Popen([launcher, "--listen", "--trajectory", path],
+                            stdin=subprocess.PIPE, stdout=subprocess.PIPE,
+                            cwd=str(tmp_path))
+    try:
+        proc.stdin.write(struct.pack("<Q", len(msg)) + msg)
+        proc.stdin.flush()
+        (size,) = struct.unpack("<Q", proc.stdout.read(8))
+        res = msgpack.unpackb(proc.stdout.read(size), raw=False)
+        assert res["i_frame"] == 1 and res["n_frames"] == 2
+        assert res["time"] == pytest.approx(0.2)
+        assert res["streamlines"] == [] and res["vortexlines"] == []
+        proc.stdin.write(struct.pack("<Q", 0))  # terminate (listener.cpp:110)
+        proc.stdin.flush()
+        assert proc.wait(timeout=30) == 0
+    finally:
+        if proc.poll() is None:
+            proc.kill()
