@@ -103,3 +103,50 @@ def test_own_precompute_periphery_operator():
     resid = np.abs(u_in - U[None, :]).max()
     assert resid < 5e-4, resid  # reference-RBF fixture achieved ~1e-6;
     # Voronoi weights are first-order accurate per cell, hence looser
+
+
+def test_surface_of_revolution_normals_weights():
+    """Analytic oocyte normals/weights
+    (precompute.surface_of_revolution_normals_weights; the operator-level
+    comparison lives in tools/check_oocyte_geometry.py — measured 10-45x
+    better interior cancellation than the reference-pipeline fixture)."""
+    import os
+    from skellysim_amd.precompute import (
+        surface_of_revolution_normals_weights, envelope_height)
+    HERE = os.path.dirname(os.path.abspath(__file__))
+    fx = np.load(os.path.join(HERE, "golden", "oocyte_nodes.npz"))
+    nodes = fx["nodes"]
+    s = float(fx["scale_factor"])
+    T, p1, p2, L = (float(fx[k]) for k in
+                    ("envelope_T", "envelope_p1", "envelope_p2",
+                     "envelope_length"))
+    g = surface_of_revolution_normals_weights(nodes, T, p1, p2, L,
+                                              scale_factor=s)
+    n, w = g["normals"], g["weights"]
+    assert np.isfinite(n).all() and np.isfinite(w).all()
+    assert np.allclose(np.linalg.norm(n, axis=1), 1.0, atol=1e-12)
+    assert np.all(w > 0)
+    # total area within 1% of the reference RBF quadrature's
+    ref_area = float(fx["quadrature_weights"].sum())
+    assert abs(w.sum() - ref_area) / ref_area < 0.01
+    # inward: normals point against the radial direction in (y, z)
+    rad = np.linalg.norm(nodes[:, 1:], axis=1)
+    mid = rad > 0.2
+    assert np.all((n[mid, 1] * nodes[mid, 1] + n[mid, 2] * nodes[mid, 2]) < 0)
+    # orthogonal to the numeric meridian tangent away from the caps
+    x0 = nodes[:, 0] / s
+    sel = np.abs(2 * x0 / L) < 0.8
+    eps = 1e-6
+    hp, _ = envelope_height(x0[sel] + eps, T, p1, p2, L)
+    hm, _ = envelope_height(x0[sel] - eps, T, p1, p2, L)
+    dH = (hp - hm) / (2 * eps)  # h'(x/s) = dH/dx in scaled coords
+    theta = np.arctan2(nodes[sel, 2], nodes[sel, 1])
+    tang = np.stack([np.ones(int(sel.sum())), dH * np.cos(theta),
+                     dH * np.sin(theta)], axis=1)
+    tang /= np.linalg.norm(tang, axis=1)[:, None]
+    assert np.abs(np.einsum("ni,ni->n", n[sel], tang)).max() < 1e-4
+    # agreement with the fixture normals away from the caps (the fixture
+    # degrades near the tips; see tools/check_oocyte_geometry.py)
+    ref_n = fx["normals"] / np.linalg.norm(fx["normals"], axis=1)[:, None]
+    cosang = np.einsum("ni,ni->n", n[sel], ref_n[sel])
+    assert np.quantile(cosang, 0.05) > 0.999
