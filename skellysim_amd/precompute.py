@@ -95,6 +95,42 @@ def surface_of_revolution_normals_weights(nodes, T, p1, p2, length,
     return dict(normals=n, weights=vertex_area_weights(nodes))
 
 
+def surface_of_revolution_geometry(n_nodes, T, p1, p2, length,
+                                   scale_factor=1.0, inward_normals=True):
+    """Self-contained oocyte-style SOR geometry: nodes placed by inverting
+    the cumulative surface area along the axis (equal-area spacing, so node
+    density follows the local area element h sqrt(1+h'^2)) with
+    golden-angle azimuths (spiral lattice, like the Fibonacci sphere);
+    normals analytic, weights = vertex areas. Returns dict(nodes, normals,
+    weights) in the SCALED frame (the reference scales nodes by
+    scale_factor, precompute.py:34).
+
+    Quality note: the spiral lattice on this elongated surface yields
+    boundary operators with interior cancellation ~5e-3 at 500-2000 nodes
+    (skinny hull triangles limit the vertex-area weights). For production
+    oocyte work prefer the reference shape-gallery NODE SET combined with
+    surface_of_revolution_normals_weights (8.9e-5 at 6431 nodes —
+    tools/check_oocyte_geometry.py)."""
+    from scipy.integrate import cumulative_trapezoid
+    xs = np.linspace(-length / 2, length / 2, 20001)[1:-1]
+    h, dh = envelope_height(xs, T, p1, p2, length)
+    darea = h * np.sqrt(1.0 + dh * dh)
+    cum = np.concatenate([[0.0], cumulative_trapezoid(darea, xs)])
+    cum /= cum[-1]
+    frac = (np.arange(n_nodes) + 0.5) / n_nodes
+    x_nodes = np.interp(frac, cum, xs)
+    hx, _ = envelope_height(x_nodes, T, p1, p2, length)
+    theta = np.pi * (1.0 + 5.0 ** 0.5) * np.arange(n_nodes)
+    nodes = scale_factor * np.stack(
+        [x_nodes, hx * np.cos(theta), hx * np.sin(theta)], axis=1)
+    nw = surface_of_revolution_normals_weights(
+        nodes, T, p1, p2, length, scale_factor=scale_factor,
+        inward_normals=inward_normals)
+    return dict(nodes=nodes, normals=nw["normals"], weights=nw["weights"],
+                envelope_T=T, envelope_p1=p1, envelope_p2=p2,
+                envelope_length=length, scale_factor=scale_factor)
+
+
 def ellipsoid_geometry(n_nodes, a, b, c, inward_normals=False):
     """dict(nodes, normals, weights) for the ellipsoid
     (x/a)^2+(y/b)^2+(z/c)^2=1: unit-sphere Fibonacci nodes mapped by
