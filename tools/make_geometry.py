@@ -28,7 +28,8 @@ from skellysim_amd.precompute import (sphere_geometry, ellipsoid_geometry,
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("shape", choices=["sphere", "ellipsoid"])
+    ap.add_argument("shape", choices=["sphere", "ellipsoid",
+                                      "surface_of_revolution"])
     ap.add_argument("--n-nodes", type=int, required=True)
     ap.add_argument("--radius", type=float, default=1.0)
     ap.add_argument("--abc", type=float, nargs=3, default=None)
@@ -38,20 +39,33 @@ def main():
     ap.add_argument("--attachment-scale", type=float, default=1.1,
                     help="site radius / surface radius (sites must sit OFF "
                          "the quadrature surface)")
+    ap.add_argument("--envelope", type=float, nargs=4, default=None,
+                    metavar=("T", "P1", "P2", "LENGTH"),
+                    help="surface_of_revolution envelope parameters "
+                         "(oocyte defaults: 0.72 0.4 0.2 7.5)")
+    ap.add_argument("--scale-factor", type=float, default=1.04)
     ap.add_argument("--out", required=True)
     args = ap.parse_args()
 
     if args.shape == "sphere":
         g = sphere_geometry(args.n_nodes, args.radius,
                             inward_normals=args.inward)
-    else:
+    elif args.shape == "ellipsoid":
         if args.abc is None:
             ap.error("ellipsoid requires --abc a b c")
         g = ellipsoid_geometry(args.n_nodes, *args.abc,
                                inward_normals=args.inward)
+    else:
+        from skellysim_amd.precompute import surface_of_revolution_geometry
+        if args.envelope is None:
+            ap.error("surface_of_revolution requires --envelope T p1 p2 L")
+        g = surface_of_revolution_geometry(args.n_nodes, *args.envelope,
+                                           scale_factor=args.scale_factor,
+                                           inward_normals=args.inward)
     out = dict(nodes=g["nodes"], normals=g["normals"], weights=g["weights"],
                quadrature_weights=g["weights"])
-    for k in ("radius", "a", "b", "c"):
+    for k in ("radius", "a", "b", "c", "envelope_T", "envelope_p1",
+              "envelope_p2", "envelope_length", "scale_factor"):
         if k in g:
             out[k] = g[k]
     if args.nucleation_sites:
