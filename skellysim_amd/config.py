@@ -156,6 +156,8 @@ def build_system(cfg, backend=None, shell_geometry=None, dt=None,
                       periphery_binding=params.get("periphery_binding"),
                       dynamic_instability=di,
                       seed=params.get("seed", 130319))
+    system.motor_activation_delay = params.get(
+        "implicit_motor_activation_delay", 0.0)
     # point/background sources (system.cpp:691-695)
     from .sources import PointSourceContainer, BackgroundSource
     if cfg.get("point_sources"):

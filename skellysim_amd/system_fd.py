@@ -133,6 +133,9 @@ class SystemFD:
         self.point_sources = None
         self.background_source = None
         self.time = 0.0
+        # motor forces held at zero until the clock passes this
+        # (params.implicit_motor_activation_delay, system.cpp:417-419)
+        self.motor_activation_delay = 0.0
         self.background_flow = background_flow  # fn: (n,3) -> (n,3)
         # steric fiber-periphery repulsion (system.cpp:421, params.cpp:18):
         # dict(kind="sphere"|"ellipsoid", f_0=, l_0=, radius=|abc=) or None
@@ -312,10 +315,12 @@ class SystemFD:
         r_all = self.all_nodes()
         nf_nodes = self.fiber_node_count
 
-        # motor force (generate_constant_force, f_c_fd.cpp:160-169)
+        # motor force (generate_constant_force, f_c_fd.cpp:160-169); held
+        # at zero until the activation delay passes (system.cpp:417-419)
         motor = np.zeros((nf_nodes, 3))
-        for f, a, b in self._fiber_node_slices():
-            motor[a:b] = (f.force_scale * f.xs).T
+        if not (self.motor_activation_delay > self.time):
+            for f, a, b in self._fiber_node_slices():
+                motor[a:b] = (f.force_scale * f.xs).T
 
         # fiber-periphery steric repulsion (fc_->periphery_force,
         # system.cpp:421; per-fiber force periphery.cpp:140-162,232-263)

@@ -247,3 +247,29 @@ def test_two_fiber_system_converges_and_is_finite():
         seg = np.diff(f.x.T, axis=0)
         length = np.linalg.norm(seg, axis=1).sum()
         assert abs(length - f.length) / f.length < 5e-3
+
+
+def test_motor_activation_delay():
+    """Motor forces stay off until the clock passes
+    implicit_motor_activation_delay (system.cpp:417-419): a straight
+    motor-forced fiber (no bending forces) is quiescent before the delay
+    and translates after."""
+    def make():
+        fib = straight_fiber(n=16, force_scale=-0.05)
+        sys_ = SystemFD([fib], eta=1.0, dt=0.05, backend=OracleBackend())
+        sys_.motor_activation_delay = 0.2
+        return sys_, fib
+
+    sys_, fib = make()
+    x0 = fib.x.copy()
+    sys_.time = 0.0
+    assert sys_.step(tol=1e-12, maxiter=200)["converged"]
+    moved_before = np.abs(fib.x - x0).max()
+    assert moved_before < 1e-10  # motor off, straight fiber -> no motion
+
+    sys2, fib2 = make()
+    x0b = fib2.x.copy()
+    sys2.time = 0.3          # past the delay
+    assert sys2.step(tol=1e-12, maxiter=200)["converged"]
+    moved_after = np.abs(fib2.x - x0b).max()
+    assert moved_after > 1e-4
