@@ -532,16 +532,52 @@ class SystemFD:
         from .batched import BatchedLU
         d["lu"] = BatchedLU(d["A"])
 
+        if self.bodies:
+            # per-body blocks + fiber<->body link statics (body.py)
+            blocks = []
+            for b in self.bodies:
+                A_lu = torch.linalg.lu_factor(T(b._A_dense))
+                blocks.append(dict(n=b.n_nodes,
+                                   e=T(np.stack(b.e_sub)),     # (3, n, 3)
+                                   w=T(b.weights), K=T(b.K), lu=A_lu))
+            d["bodies"] = blocks
+            d["body_nodes"] = T(self.body_nodes())
+            d["body_normals"] = T(self.body_normals())
+            d["centers"] = T(np.stack([b.position for b in self.bodies]))
+            att = [(i, f.binding_site) for i, f in enumerate(self.fibers)
+                   if f.binding_site[0] >= 0]
+            d["att_idx"] = torch.tensor([i for i, _ in att], dtype=torch.long,
+                                        device=dev)
+            d["att_body"] = torch.tensor([bs[0] for _, bs in att],
+                                         dtype=torch.long, device=dev)
+            site = np.stack([self.bodies[bs[0]].nucleation_sites[bs[1]]
+                             - self.bodies[bs[0]].position
+                             for _, bs in att]) if att else np.zeros((0, 3))
+            d["att_site"] = T(site)
+            d["att_site_hat"] = T(site / np.linalg.norm(site, axis=1,
+                                                        keepdims=True)) \
+                if len(att) else T(site)
+            d["d2c0"] = T(f0.mats["D_2_0"][:, 0])
+            d["d3c0"] = T(f0.mats["D_3_0"][:, 0])
+            d["E_fib"] = T(np.array([f.bending_rigidity for f in self.fibers]))
+            d["s2"] = T(np.array([(2.0 / f.length) ** 2 for f in self.fibers]))
+            d["s3"] = T(np.array([(2.0 / f.length) ** 3 for f in self.fibers]))
+
     def _apply_matvec_device(self, x):
-        """apply_matvec entirely on device (torch fp64 CUDA vector in/out)."""
+        """apply_matvec entirely on device (torch fp64 CUDA vector in/out);
+        covers fibers + shell + bodies (node order [fib | shell | body],
+        system.cpp:269-324)."""
         import torch
         from .evaluator import stokeslet_device, stresslet_device
         d = self._dev
         nf, n = d["nf"], d["n"]
         eta = self.eta
         nf_nodes = nf * n
+        sh_size = self.shell_sol_size
+        sh_nodes_n = sh_size // 3
         x_fib = x[: 4 * nf_nodes].reshape(nf, 4 * n)
-        x_shell = x[4 * nf_nodes:]
+        x_shell = x[4 * nf_nodes: 4 * nf_nodes + sh_size]
+        x_bodies = x[4 * nf_nodes + sh_size:]
 
         # forces: F @ x (component-major) -> node-major (nf*n, 3)
         fw = torch.bmm(d["F"], x_fib.unsqueeze(-1)).squeeze(-1)     # (nf, 3n)
@@ -555,22 +591,101 @@ class SystemFD:
         v_all[:nf_nodes] -= corr
 
         if self.shell:
+            # shell double layer flows to fibers and bodies, not to itself
             dens = x_shell.reshape(-1, 3)
             f_dl = 2.0 * eta * torch.einsum("ni,nj->nij", d["sh_normals"],
                                             dens).reshape(-1, 9).contiguous()
             v_all[:nf_nodes] += stresslet_device(d["sh_nodes"], f_dl,
                                                  d["r_fib"], eta)
+            if self.bodies:
+                v_all[nf_nodes + sh_nodes_n:] += stresslet_device(
+                    d["sh_nodes"], f_dl, d["body_nodes"], eta)
+
+        vel7 = None
+        if self.bodies:
+            from .evaluator import rotlet_device
+            nb = len(self.bodies)
+            # per-body [U, w] slices of the body block
+            body_vels = torch.zeros((nb, 6), dtype=x.dtype, device=x.device)
+            dens_parts, off = [], 0
+            for bi, bd in enumerate(d["bodies"]):
+                nn = bd["n"]
+                dens_parts.append(x_bodies[off: off + 3 * nn].reshape(nn, 3))
+                body_vels[bi] = x_bodies[off + 3 * nn: off + 3 * nn + 6]
+                off += 3 * nn + 6
+            bdens = torch.cat(dens_parts)
+
+            # link conditions (body_container.cpp:171-268) in torch
+            body_ft = torch.zeros((nb, 6), dtype=x.dtype, device=x.device)
+            ai = d["att_idx"]
+            if len(ai):
+                xa = x_fib[ai]
+                x_new = xa[:, : 3 * n].reshape(-1, 3, n)
+                T0 = xa[:, 3 * n]
+                xss0 = torch.einsum("ain,n->ai", x_new, d["d2c0"]) \
+                    * d["s2"][ai][:, None]
+                xsss0 = torch.einsum("ain,n->ai", x_new, d["d3c0"]) \
+                    * d["s3"][ai][:, None]
+                xs0 = d["xs"][ai][:, :, 0]
+                E = d["E_fib"][ai][:, None]
+                site = d["att_site"]
+                F_b = -E * xsss0 + xs0 * T0[:, None]
+                L_b = (-E * torch.cross(site, xsss0, dim=1)
+                       + torch.cross(site, xs0, dim=1) * T0[:, None]
+                       + E * torch.cross(xs0, xss0, dim=1))
+                body_ft.index_add_(0, d["att_body"],
+                                   torch.cat([F_b, L_b], dim=1))
+                U_att = body_vels[d["att_body"], 0:3]
+                w_att = body_vels[d["att_body"], 3:6]
+                v_f = -U_att - torch.cross(w_att, site, dim=1)
+                tc = -(xs0 * U_att).sum(dim=1) \
+                    + (torch.cross(xs0, site, dim=1) * w_att).sum(dim=1)
+                w_f = torch.cross(d["att_site_hat"], w_att, dim=1)
+                vel7 = torch.zeros((nf, 7), dtype=x.dtype, device=x.device)
+                vel7[ai] = torch.cat([v_f, tc[:, None], w_f], dim=1)
+
+            # body flow: double layer + center stokeslet/rotlet of link F/T
+            f_dl_b = 2.0 * eta * torch.einsum(
+                "ni,nj->nij", d["body_normals"], bdens).reshape(-1, 9).contiguous()
+            v_all += stresslet_device(d["body_nodes"], f_dl_b, d["r_all"], eta)
+            v_all += stokeslet_device(d["centers"],
+                                      body_ft[:, 0:3].contiguous(),
+                                      d["r_all"], eta)
+            v_all += rotlet_device(d["centers"], d["r_all"],
+                                   body_ft[:, 3:6].contiguous(), eta)
 
         res = torch.empty_like(x)
-        res[: 4 * nf_nodes] = self._fiber_block_device(x_fib, v_all[:nf_nodes]).reshape(-1)
+        res[: 4 * nf_nodes] = self._fiber_block_device(
+            x_fib, v_all[:nf_nodes], vel7).reshape(-1)
         if self.shell:
-            v_shell = v_all[nf_nodes:].reshape(-1)
-            res[4 * nf_nodes:] = d["sh_A"] @ x_shell + v_shell
+            v_shell = v_all[nf_nodes: nf_nodes + sh_nodes_n].reshape(-1)
+            res[4 * nf_nodes: 4 * nf_nodes + sh_size] = \
+                d["sh_A"] @ x_shell + v_shell
+        if self.bodies:
+            off = 0
+            off_node = nf_nodes + sh_nodes_n
+            base = 4 * nf_nodes + sh_size
+            for bd in d["bodies"]:
+                nn = bd["n"]
+                xb = x_bodies[off: off + 3 * nn + 6]
+                dloc = xb[: 3 * nn].reshape(nn, 3)
+                U = xb[3 * nn:]
+                sub = (dloc[:, 0:1] * bd["e"][0] + dloc[:, 1:2] * bd["e"][1]
+                       + dloc[:, 2:3] * bd["e"][2]) / bd["w"][:, None]
+                v_nodes = v_all[off_node: off_node + nn]
+                res[base + off: base + off + 3 * nn] = \
+                    (-sub + v_nodes).reshape(-1) - bd["K"] @ U
+                res[base + off + 3 * nn: base + off + 3 * nn + 6] = \
+                    -bd["K"].T @ xb[: 3 * nn] + U
+                off += 3 * nn + 6
+                off_node += nn
         return res
 
-    def _fiber_block_device(self, x_fib, v_fib_nodes):
+    def _fiber_block_device(self, x_fib, v_fib_nodes, vel7=None):
         """Per-fiber operator block: A x - vT_in + the two BC velocity
-        corrections (fiber_fd.matvec, fiber_finite_difference.cpp:278-315).
+        corrections (fiber_fd.matvec, fiber_finite_difference.cpp:278-315),
+        plus the y_BC link-condition rows when vel7 (nf, 7) is given
+        (fiber matvec's v_boundary, f_c_fd.cpp:226).
         x_fib (nf, 4n), v_fib_nodes (nf*n, 3) node-major -> (nf, 4n)."""
         import torch
         d = self._dev
@@ -588,22 +703,37 @@ class SystemFD:
         res_fib[:, bc_start + 3] += (v_fib[:, :, 0] * d["xs"][:, :, 0]).sum(dim=1)
         res_fib[:, bc_start + 10] += d["plus_vel"] * \
             (v_fib[:, :, -1] * d["xs"][:, :, -1]).sum(dim=1)
+        if vel7 is not None:
+            res_fib[:, bc_start: bc_start + 7] += vel7
         return res_fib
 
     def _apply_precond_device(self, x):
         import torch
         d = self._dev
         nf, n = d["nf"], d["n"]
+        sh_size = self.shell_sol_size
         res = torch.empty_like(x)
         res[: 4 * nf * n] = d["lu"].solve(x[: 4 * nf * n].reshape(nf, 4 * n)).reshape(-1)
         if self.shell:
-            res[4 * nf * n:] = d["sh_Minv"] @ x[4 * nf * n:]
+            res[4 * nf * n: 4 * nf * n + sh_size] = \
+                d["sh_Minv"] @ x[4 * nf * n: 4 * nf * n + sh_size]
+        if self.bodies:
+            off = 4 * nf * n + sh_size
+            for bd in d["bodies"]:
+                m = 3 * bd["n"] + 6
+                res[off: off + m] = torch.linalg.lu_solve(
+                    bd["lu"][0], bd["lu"][1],
+                    x[off: off + m].unsqueeze(-1)).squeeze(-1)
+                off += m
         return res
 
-    def solve(self, tol=1e-10, maxiter=200, restart=None):
+    def solve(self, tol=1e-10, maxiter=200, restart=None, device_mode=None):
         """system.cpp:464-478 via the engine GMRES (right-preconditioned,
         ICGS — solver_hydro.cpp:64-87). With uniform fibers on the HIP
-        backend the whole iteration runs device-resident."""
+        backend the whole iteration runs device-resident. device_mode=None
+        auto-selects (host path when bodies are present — their device
+        blocks exist but are validated on CPU tensors only this round;
+        pass device_mode=True to opt in)."""
         import torch
         from .gmres import gmres
 
@@ -611,15 +741,17 @@ class SystemFD:
         if restart is None:
             restart = min(200, maxiter)
 
-        device_mode = (self._uniform and self.fibers and not self.bodies
-                       and isinstance(self.backend, HipBackend))
+        if device_mode is None:
+            device_mode = (self._uniform and self.fibers and not self.bodies
+                           and isinstance(self.backend, HipBackend))
         if device_mode:
             self._build_device_operators()
             b = self.backend._t(rhs)
             x, info = gmres(self._apply_matvec_device, b,
                             precond=self._apply_precond_device,
                             tol=tol, maxiter=maxiter, restart=restart)
-            self.backend.torch.cuda.synchronize()
+            if b.is_cuda:
+                self.backend.torch.cuda.synchronize()
             self.solution = x.cpu().numpy()
             return info
 

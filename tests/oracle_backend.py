@@ -70,10 +70,19 @@ class FakeDeviceBackend(OracleBackend):
 def patch_device_kernels_with_oracle():
     """Replace skellysim_amd.evaluator's device kernels with oracle-backed
     CPU-tensor fakes (identical math by construction). TEST-ONLY; call in a
-    worker subprocess, never in product code."""
+    worker subprocess (or restore() in-process). Returns a restore
+    callable."""
     import torch
     import oracle
     import skellysim_amd.evaluator as ev
+
+    saved = {k: getattr(ev, k) for k in
+             ("stokeslet_device", "stresslet_device",
+              "oseen_tensor_batched_device", "rotlet_device")}
+
+    def restore():
+        for k, v in saved.items():
+            setattr(ev, k, v)
 
     def _np(t):
         return np.ascontiguousarray(t.detach().cpu().numpy())
@@ -84,5 +93,8 @@ def patch_device_kernels_with_oracle():
         oracle.stresslet(_np(r), _np(f), _np(t), eta))
     ev.oseen_tensor_batched_device = lambda pts, eta=1.0: torch.from_numpy(
         np.stack([oracle.oseen_tensor(p, eta) for p in _np(pts)]))
+    ev.rotlet_device = lambda r, t, rho, eta=1.0, *a, **k: torch.from_numpy(
+        oracle.rotlet(_np(r), _np(t), _np(rho), eta))
+    return restore
 
 
