@@ -69,6 +69,22 @@ def periphery_interaction_from(cfg):
     return None
 
 
+def _geometry_fields(g):
+    """Accept both this engine's key names (nodes/normals/weights|
+    quadrature_weights) and the reference precompute's
+    (node_positions_ref/node_normals_ref/node_weights,
+    precompute.py:187,245 / body_spherical.cpp:199-203)."""
+    def pick(*names):
+        for k in names:
+            if k in g:
+                return np.asarray(g[k], float)
+        raise KeyError(f"none of {names} in geometry ({list(g.keys())})")
+    nodes = pick("nodes", "node_positions_ref")
+    normals = pick("normals", "node_normals_ref")
+    weights = pick("weights", "quadrature_weights", "node_weights")
+    return nodes, normals, weights.reshape(-1)
+
+
 def build_bodies(cfg, body_geometry):
     """Spherical bodies from reference [[bodies]] tables
     (skelly_config.py:720-751 / body_spherical.cpp:213-275). body_geometry:
@@ -103,9 +119,7 @@ def build_bodies(cfg, body_geometry):
                       "external_oscillation_force_frequency", 0.0),
                   oscillation_phase=bt.get(
                       "external_oscillation_force_phase", 0.0))
-        nodes = g["nodes"]
-        normals = g["normals"]
-        w = np.asarray(g["weights"]).reshape(-1)
+        nodes, normals, w = _geometry_fields(g)
         if shape == "sphere":
             bodies.append(SphericalBody(nodes, normals, w,
                                         bt.get("radius", 1.0), **kw))
@@ -137,11 +151,19 @@ def build_system(cfg, backend=None, shell_geometry=None, dt=None,
         g = np.load(shell_geometry) if isinstance(shell_geometry, str) \
             else shell_geometry
         dev = torch.device("cuda:0")
-        A, M_inv = assemble_shell_operator(
-            torch.from_numpy(np.asarray(g["nodes"])).to(dev),
-            torch.from_numpy(np.asarray(g["normals"])).to(dev),
-            torch.from_numpy(np.asarray(g["quadrature_weights"])).to(dev))
-        shell = Shell(np.asarray(g["nodes"]), np.asarray(g["normals"]), A, M_inv)
+        nodes, normals, weights = _geometry_fields(g)
+        if "stresslet_plus_complementary" in g and "M_inv" in g:
+            # a full reference precompute file (precompute.py:141-148):
+            # adopt its prebuilt dense operators directly
+            A = torch.from_numpy(
+                np.asarray(g["stresslet_plus_complementary"])).to(dev)
+            M_inv = torch.from_numpy(np.asarray(g["M_inv"])).to(dev)
+        else:
+            A, M_inv = assemble_shell_operator(
+                torch.from_numpy(nodes).to(dev),
+                torch.from_numpy(normals).to(dev),
+                torch.from_numpy(weights).to(dev))
+        shell = Shell(nodes, normals, A, M_inv)
 
     bodies = build_bodies(cfg, body_geometry) if body_geometry is not None \
         else []

@@ -256,3 +256,26 @@ nucleation_sites = [{sites_flat}]
     assert max(counts) <= 6          # bounded by the sites
     for f in traj.frames:            # bodies recorded every frame
         assert len(f["bodies"][0]) == 1
+
+
+def test_reference_precompute_key_aliases():
+    """Geometry loaders accept the reference precompute's own key names
+    (node_positions_ref/node_normals_ref/node_weights for bodies,
+    precompute.py:187; nodes/normals/quadrature_weights(+operators) for
+    peripheries, precompute.py:141-148)."""
+    from skellysim_amd.config import build_bodies, _geometry_fields
+    fx = np.load(os.path.join(HERE, "golden", "periphery_sphere_192.npz"))
+    R = float(fx["radius"])
+    ref_style = {"node_positions_ref": fx["nodes"],
+                 "node_normals_ref": -fx["normals"],
+                 "node_weights": fx["quadrature_weights"]}
+    nodes, normals, w = _geometry_fields(ref_style)
+    assert nodes.shape == (192, 3) and w.shape == (192,)
+    cfg = {"bodies": [dict(shape="sphere", radius=R)]}
+    (b,) = build_bodies(cfg, ref_style)
+    assert b.n_nodes == 192 and b.radius == R
+    # engine-style keys still work
+    eng_style = {"nodes": fx["nodes"], "normals": -fx["normals"],
+                 "weights": fx["quadrature_weights"]}
+    (b2,) = build_bodies(cfg, eng_style)
+    assert np.allclose(b2.nodes, b.nodes)
