@@ -361,13 +361,17 @@ def main():
     compute = HipBackend()
     shell_geometry = None
     if args.shell_geometry:
+        from .config import _geometry_fields
         fx = np.load(args.shell_geometry)
-        shell_geometry = {"nodes": fx["nodes"], "normals": fx["normals"]}
+        nodes, normals, _ = _geometry_fields(fx)
+        shell_geometry = {"nodes": nodes, "normals": normals}
     body_geometry = None
     if args.body_geometry:
+        from .config import _geometry_fields
         bx = np.load(args.body_geometry)
-        body_geometry = {"nodes": bx["nodes"], "normals": bx["normals"],
-                         "weights": bx["weights"]}
+        nodes, normals, weights = _geometry_fields(bx)
+        body_geometry = {"nodes": nodes, "normals": normals,
+                         "weights": weights}
         if "nucleation_sites" in bx:
             body_geometry["nucleation_sites"] = bx["nucleation_sites"]
     sources = None
