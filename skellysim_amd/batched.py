@@ -10,6 +10,7 @@ rocSOLVER batched getrf/getrs via torch.linalg.lu_factor / lu_solve with all
 factors resident in HBM across the solve.
 """
 
+import numpy as np
 import torch
 
 _magma_latched = False
@@ -144,6 +145,29 @@ class BatchedLU:
                     torch.linalg.vector_norm(v) / max(1, mats.shape[0]) ** 0.5)
         return math.isfinite(rel) and rel < rtol
 
+    @staticmethod
+    def _use_trsm():
+        """SKELLY_LU_TRSM=1 routes the per-iteration solves through batched
+        solve_triangular (rocBLAS trsm) instead of torch.linalg.lu_solve
+        (magma batched trsv on ROCm). Round-2 experiment knob: magma's
+        batched calls build per-call device pointer arrays, a use-after-free
+        candidate for the deep-queue (sync-cadence) corruption documented
+        in gmres.py."""
+        import os
+        return os.environ.get("SKELLY_LU_TRSM", "0") == "1"
+
+    def _perm_from_pivots(self):
+        if getattr(self, "_perm", None) is None:
+            piv = self.pivots.cpu().numpy() - 1      # LAPACK 1-based
+            nb, m = piv.shape
+            perm = np.tile(np.arange(m), (nb, 1))
+            for b in range(nb):
+                for i in range(m):
+                    j = piv[b, i]
+                    perm[b, [i, j]] = perm[b, [j, i]]
+            self._perm = torch.from_numpy(perm).to(self.LU.device)
+        return self._perm
+
     def solve(self, rhs):
         """rhs: (n_fibers, m) or (n_fibers, m, k) -> same shape solution."""
         if self._host_lus is not None:
@@ -156,7 +180,15 @@ class BatchedLU:
         squeeze = rhs.dim() == 2
         if squeeze:
             rhs = rhs.unsqueeze(-1)
-        x = torch.linalg.lu_solve(self.LU, self.pivots, rhs)
+        if self._use_trsm():
+            perm = self._perm_from_pivots()          # (nb, m)
+            b = torch.gather(rhs, 1,
+                             perm.unsqueeze(-1).expand(-1, -1, rhs.shape[-1]))
+            y = torch.linalg.solve_triangular(self.LU, b, upper=False,
+                                              unitriangular=True)
+            x = torch.linalg.solve_triangular(self.LU, y, upper=True)
+        else:
+            x = torch.linalg.lu_solve(self.LU, self.pivots, rhs)
         return x.squeeze(-1) if squeeze else x
 
 

@@ -46,3 +46,21 @@ def test_shape_and_dtype_validation():
         BatchedLU(torch.zeros(3, 4, 5, dtype=torch.float64))
     with pytest.raises(TypeError):
         BatchedLU(torch.zeros(3, 4, 4, dtype=torch.float32))
+
+
+def test_lu_trsm_path_matches_lu_solve(monkeypatch):
+    """SKELLY_LU_TRSM=1 (round-2 experiment: batched solve_triangular
+    instead of magma lu_solve) must produce the same solutions."""
+    import torch
+    from skellysim_amd.batched import BatchedLU
+    rng = np.random.default_rng(3)
+    mats = torch.from_numpy(rng.uniform(-1, 1, (5, 24, 24))
+                            + 24 * np.eye(24)[None])
+    rhs = torch.from_numpy(rng.uniform(-1, 1, (5, 24)))
+    lu = BatchedLU(mats.clone())
+    x_ref = lu.solve(rhs)
+    monkeypatch.setenv("SKELLY_LU_TRSM", "1")
+    x_trsm = lu.solve(rhs)
+    assert torch.allclose(x_trsm, x_ref, atol=1e-12)
+    direct = torch.linalg.solve(mats, rhs.unsqueeze(-1)).squeeze(-1)
+    assert torch.allclose(x_trsm, direct, atol=1e-10)
