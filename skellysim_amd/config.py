@@ -134,7 +134,7 @@ def build_bodies(cfg, body_geometry):
 
 
 def build_system(cfg, backend=None, shell_geometry=None, dt=None,
-                 background_flow=None, body_geometry=None):
+                 background_flow=None, body_geometry=None, device=None):
     """SystemFD from a reference config. shell_geometry: npz path or dict
     with nodes/normals/quadrature_weights (the periphery geometry the
     reference's precompute generates; operators are assembled on device)."""
@@ -150,7 +150,7 @@ def build_system(cfg, backend=None, shell_geometry=None, dt=None,
 
         g = np.load(shell_geometry) if isinstance(shell_geometry, str) \
             else shell_geometry
-        dev = torch.device("cuda:0")
+        dev = torch.device(device if device is not None else "cuda:0")
         nodes, normals, weights = _geometry_fields(g)
         if "stresslet_plus_complementary" in g and "M_inv" in g:
             # a full reference precompute file (precompute.py:141-148):

@@ -117,6 +117,7 @@ def test_body_config_build_and_solve(tmp_path):
     from oracle_backend import OracleBackend
     from skellysim_amd.config import load_config, build_bodies, build_system
 
+    from oracle_backend import OracleBackend
     fx = np.load(os.path.join(HERE, "golden", "periphery_sphere_192.npz"))
     R = float(fx["radius"])
     geom = {"nodes": fx["nodes"], "normals": -fx["normals"],
@@ -264,6 +265,7 @@ def test_reference_precompute_key_aliases():
     precompute.py:187; nodes/normals/quadrature_weights(+operators) for
     peripheries, precompute.py:141-148)."""
     from skellysim_amd.config import build_bodies, _geometry_fields
+    from oracle_backend import OracleBackend
     fx = np.load(os.path.join(HERE, "golden", "periphery_sphere_192.npz"))
     R = float(fx["radius"])
     ref_style = {"node_positions_ref": fx["nodes"],
@@ -279,3 +281,23 @@ def test_reference_precompute_key_aliases():
                  "weights": fx["quadrature_weights"]}
     (b2,) = build_bodies(cfg, eng_style)
     assert np.allclose(b2.nodes, b.nodes)
+
+
+def test_shell_precompute_operator_adoption():
+    """A full reference precompute npz (with stresslet_plus_complementary
+    and M_inv) is adopted directly as the shell operators instead of
+    re-assembling (precompute.py:141-148 file layout)."""
+    import torch
+    from skellysim_amd.config import build_system
+    from oracle_backend import OracleBackend
+    fx = np.load(os.path.join(HERE, "golden", "periphery_sphere_192.npz"))
+    cfg = {"params": {"eta": 1.0}, "periphery": {"shape": "sphere",
+                                                 "radius": float(fx["radius"])},
+           "fibers": []}
+    sys_ = build_system(cfg, backend=OracleBackend(),
+                        shell_geometry=dict(fx), device="cpu")
+    assert sys_.shell is not None
+    assert torch.is_tensor(sys_.shell.A)
+    assert np.allclose(sys_.shell.A.numpy(),
+                       fx["stresslet_plus_complementary"])
+    assert np.allclose(sys_.shell.M_inv.numpy(), fx["M_inv"])
