@@ -107,7 +107,11 @@ class TrajectoryWriter:
         frame = {
             "time": float(time),
             "dt": float(dt),
-            "rng_state": [],
+            # the reference stores its Philox counters here
+            # (io_maps.hpp:31-41); this engine stores its numpy PCG64 state
+            # (128-bit ints as decimal strings, msgpack has no int128) so a
+            # resumed dynamic-instability run continues the same stream
+            "rng_state": _pack_rng(getattr(system, "rng", None)),
             "fibers": [FIBERTYPE_FINITEDIFFERENCE,
                        [fiber_map(f) for f in system.fibers]],
             # [spherical, deformable, ellipsoidal] (body_container.hpp:158)
@@ -128,14 +132,37 @@ class TrajectoryWriter:
         self.close()
 
 
+def _pack_rng(rng):
+    if rng is None:
+        return []
+    st = rng.bit_generator.state
+    return {"bit_generator": st["bit_generator"],
+            "state": {k: str(v) for k, v in st["state"].items()},
+            "has_uint32": int(st.get("has_uint32", 0)),
+            "uinteger": int(st.get("uinteger", 0))}
+
+
+def _restore_rng(rng, packed):
+    if rng is None or not isinstance(packed, dict):
+        return
+    if packed.get("bit_generator") != rng.bit_generator.state["bit_generator"]:
+        return
+    rng.bit_generator.state = {
+        "bit_generator": packed["bit_generator"],
+        "state": {k: int(v) for k, v in packed["state"].items()},
+        "has_uint32": int(packed.get("has_uint32", 0)),
+        "uinteger": int(packed.get("uinteger", 0))}
+
+
 def resume_from_trajectory(system, path):
     """System::resume_from_trajectory (system.cpp:223-228 +
     TrajectoryReader::unpack_current_frame): restore the system to the LAST
     frame of an existing trajectory — fibers rebuilt from their serialized
     state, body positions/orientations/solutions restored onto the
     configured bodies, the clock and dt adopted. Returns the frame count.
-    (The reference also restores its Philox rng_state; this engine's numpy
-    Generator stream is a documented deviation, see instability.py.)"""
+    The rng_state field (this engine's PCG64 state, or the reference's
+    Philox counters which are skipped) is restored so a resumed
+    dynamic-instability run continues the exact random stream."""
     from .listener import Trajectory, fibers_from_frame
     traj = Trajectory(path)
     if not traj.frames:
@@ -143,6 +170,7 @@ def resume_from_trajectory(system, path):
     frame = traj.frames[-1]
     system.time = float(frame["time"])
     system.dt = float(frame["dt"])
+    _restore_rng(getattr(system, "rng", None), frame.get("rng_state"))
     system.fibers = fibers_from_frame(frame, system.eta)
     system._uniform = all(f.n_nodes == system.fibers[0].n_nodes
                           for f in system.fibers) if system.fibers else True

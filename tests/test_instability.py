@@ -197,3 +197,45 @@ def test_di_mixed_discretization_solve():
     assert not sys_._uniform
     for f in sys_.fibers:
         assert np.isfinite(f.x).all()
+
+
+def test_di_resume_continues_random_stream(tmp_path):
+    """resume_from_trajectory restores the serialized rng state (the
+    reference restores its Philox counters the same way, system.cpp:223-228),
+    so an interrupted dynamic-instability run resumed into a fresh system
+    reproduces the uninterrupted run's nucleation/catastrophe sequence and
+    fiber states exactly."""
+    from skellysim_amd.trajectory import TrajectoryWriter, resume_from_trajectory
+
+    di = dict(n_nodes=8, v_growth=0.2, f_catastrophe=0.5,
+              nucleation_rate=10.0, min_length=0.4, radius=0.0125,
+              bending_rigidity=2.5e-3)
+
+    def fresh():
+        return SystemFD([], eta=1.0, dt=0.1, bodies=[make_body(n_sites=6)],
+                        backend=OracleBackend(), dynamic_instability=di,
+                        seed=11)
+
+    sA = fresh()
+    for _ in range(4):
+        assert sA.step(tol=1e-10, maxiter=300, restart=150)["converged"]
+        sA.time += sA.dt
+
+    path = str(tmp_path / "skelly_sim.out")
+    sB = fresh()
+    with TrajectoryWriter(path) as tw:
+        for _ in range(2):
+            assert sB.step(tol=1e-10, maxiter=300, restart=150)["converged"]
+            sB.time += sB.dt
+            tw.write_frame(sB, sB.time, sB.dt)
+    sC = fresh()
+    resume_from_trajectory(sC, path)
+    for _ in range(2):
+        assert sC.step(tol=1e-10, maxiter=300, restart=150)["converged"]
+        sC.time += sC.dt
+
+    assert len(sC.fibers) == len(sA.fibers)
+    for fA, fC in zip(sA.fibers, sC.fibers):
+        assert fC.binding_site == fA.binding_site
+        assert np.allclose(fC.x, fA.x, atol=1e-9)
+        assert fC.length == pytest.approx(fA.length, abs=1e-12)
