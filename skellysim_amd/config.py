@@ -9,6 +9,8 @@ The reference feeds the periphery matrices from its precompute .npz
 (periphery_precompute.assemble_shell_operator) from the geometry
 (nodes/normals/weights), which build_system takes as an npz path or dict."""
 
+import os
+
 import numpy as np
 import tomli
 
@@ -133,15 +135,45 @@ def build_bodies(cfg, body_geometry):
     return bodies
 
 
+def _resolve_precompute(table, config_dir):
+    """The reference loads the periphery/body surface data from the
+    precompute_file named in the TOML table (params.hpp:61 /
+    body_ellipsoidal.cpp:220-221, defaults periphery_precompute.npz /
+    body_precompute.npz, skelly_config.py:446,762). Returns a path if the
+    table names one that exists (cwd-relative like the reference, else
+    config-dir-relative), None otherwise."""
+    pf = table.get("precompute_file")
+    if not pf:
+        return None
+    if os.path.exists(pf):
+        return pf
+    if config_dir is not None:
+        cand = os.path.join(config_dir, pf)
+        if os.path.exists(cand):
+            return cand
+    return None
+
+
 def build_system(cfg, backend=None, shell_geometry=None, dt=None,
-                 background_flow=None, body_geometry=None, device=None):
+                 background_flow=None, body_geometry=None, device=None,
+                 config_dir=None):
     """SystemFD from a reference config. shell_geometry: npz path or dict
     with nodes/normals/quadrature_weights (the periphery geometry the
-    reference's precompute generates; operators are assembled on device)."""
+    reference's precompute generates; operators are assembled on device).
+    When shell_geometry/body_geometry are not supplied, the precompute_file
+    paths named in the config are picked up when they exist on disk (the
+    reference's own loading convention)."""
     params = cfg.get("params", {})
     eta = params.get("eta", 1.0)
     dt = dt if dt is not None else params.get("dt_initial", 0.025)
     fibers = build_fibers(cfg, eta)
+
+    if shell_geometry is None and "periphery" in cfg:
+        shell_geometry = _resolve_precompute(cfg["periphery"], config_dir)
+    if body_geometry is None and cfg.get("bodies"):
+        paths = [_resolve_precompute(bt, config_dir) for bt in cfg["bodies"]]
+        if all(p is not None for p in paths):
+            body_geometry = paths
 
     shell = None
     if "periphery" in cfg and shell_geometry is not None:

@@ -371,3 +371,82 @@ x = [{flat}]
     xA = np.asarray(trA.frames[-1]["fibers"][1][0]["x_"], float)
     xB = np.asarray(trB.frames[-1]["fibers"][1][0]["x_"], float)
     assert np.allclose(xB, xA, atol=1e-12)
+
+
+def test_precompute_file_auto_pickup(tmp_path):
+    """build_system honors the config's own precompute_file paths
+    (params.hpp:61 / body_ellipsoidal.cpp:220-221) when no explicit
+    geometry is supplied: the periphery operators come from the named npz
+    (adopted directly when it holds the prebuilt matrices) and each body's
+    surface from its table's file."""
+    import shutil
+    import torch
+    from oracle_backend import OracleBackend
+    from skellysim_amd.config import build_system
+
+    src = os.path.join(HERE, "golden", "periphery_sphere_192.npz")
+    shutil.copy(src, tmp_path / "periphery_precompute.npz")
+    shutil.copy(src, tmp_path / "body_precompute.npz")
+    fx = np.load(src)
+    cfg = {"params": {"eta": 1.0},
+           "periphery": {"shape": "sphere", "radius": float(fx["radius"]),
+                         "precompute_file": "periphery_precompute.npz"},
+           "bodies": [{"radius": float(fx["radius"]),
+                       "precompute_file": "body_precompute.npz"}],
+           "fibers": []}
+    sys_ = build_system(cfg, backend=OracleBackend(), device="cpu",
+                        config_dir=str(tmp_path))
+    assert sys_.shell is not None
+    assert np.allclose(sys_.shell.A.numpy(),
+                       fx["stresslet_plus_complementary"])
+    assert len(sys_.bodies) == 1
+    assert sys_.bodies[0].n_nodes == len(fx["nodes"])
+
+
+def test_skelly_sim_hip_run_dispatch(tmp_path, monkeypatch):
+    """tools/skelly_sim_hip without --listen runs the simulation with the
+    reference CLI surface (skelly_sim.cpp:27-30): --config-file=X,
+    --overwrite, --no-* forms ignored, trajectory at ./skelly_sim.out."""
+    import importlib.machinery
+    import importlib.util
+    import sys as _sys
+    from oracle_backend import OracleBackend
+
+    s = np.linspace(0, 1.0, 16)
+    pts = np.stack([0.1 * np.sin(2 * np.pi * s), np.zeros_like(s), s], axis=1)
+    flat = ", ".join(repr(float(v)) for v in pts.reshape(-1))
+    (tmp_path / "skelly_config.toml").write_text(f"""
+[params]
+eta = 1.0
+dt_initial = 0.1
+dt_write = 0.1
+t_final = 0.2
+gmres_tol = 1e-11
+adaptive_timestep_flag = false
+fiber_type = "FiniteDifference"
+
+[[fibers]]
+length = 1.0
+bending_rigidity = 2.5e-2
+radius = 0.0125
+force_scale = -0.02
+minus_clamped = false
+n_nodes = 16
+x = [{flat}]
+""")
+    loader = importlib.machinery.SourceFileLoader(
+        "skelly_sim_hip_bin",
+        os.path.join(os.path.dirname(HERE), "tools", "skelly_sim_hip"))
+    spec = importlib.util.spec_from_loader(loader.name, loader)
+    mod = importlib.util.module_from_spec(spec)
+    loader.exec_module(mod)
+    import run_sim
+    monkeypatch.setattr(run_sim, "HipBackend", OracleBackend)
+    monkeypatch.chdir(tmp_path)
+    mod.main(["--config-file=skelly_config.toml", "--overwrite",
+              "--no-resume"])
+
+    from skellysim_amd.listener import Trajectory
+    traj = Trajectory(str(tmp_path / "skelly_sim.out"))
+    assert np.allclose([f["time"] for f in traj.frames], [0.1, 0.2],
+                       atol=1e-12)

@@ -58,7 +58,9 @@ def main():
     t0 = time.perf_counter()
     sys_ = build_system(cfg, backend=HipBackend(),
                         shell_geometry=args.shell_geometry,
-                        body_geometry=args.body_geometry)
+                        body_geometry=args.body_geometry,
+                        config_dir=os.path.dirname(
+                            os.path.abspath(args.config_file)))
     print(f"system: {len(sys_.fibers)} fibers"
           + (f" + {sys_.shell.n_nodes}-node shell" if sys_.shell else "")
           + (f" + {len(sys_.bodies)} bodies" if sys_.bodies else "")
