@@ -278,3 +278,74 @@ def test_distributed_triple_matches_single_process():
     dist_sol = np.concatenate(fib_parts + shell_parts + [body_part])
     rel = np.linalg.norm(dist_sol - sys_.solution) / np.linalg.norm(sys_.solution)
     assert rel < 1e-7, rel
+
+
+def _dist_multistep_worker(rank, world, init_file, q):
+    import torch.distributed as dist
+    from skellysim_amd.system_dist import DistributedSystemFD, distribute_fibers
+    from skellysim_amd.system_fd import Shell
+    from skellysim_amd.sharded import shard_range
+    from oracle_backend import OracleBackend
+
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    try:
+        fx, fibers, bg = _make_problem()
+        N = len(fx["nodes"])
+        a, b = shard_range(N, world, rank)
+        shell = Shell(fx["nodes"], fx["normals"],
+                      fx["stresslet_plus_complementary"][3 * a: 3 * b],
+                      fx["M_inv"][3 * a: 3 * b])
+        my_fibers = distribute_fibers(fibers, world, rank)
+        sys_ = DistributedSystemFD(my_fibers, eta=1.0, dt=0.05, shell=shell,
+                                   shell_rows=(a, b), backend=OracleBackend(),
+                                   background_flow=bg)
+        for _ in range(3):
+            info = sys_.step(tol=1e-11, maxiter=300, restart=100)
+            assert info["converged"]
+        q.put((rank, [f.x for f in my_fibers]))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_distributed_multistep_matches_single_process():
+    """Three full timesteps (solve -> adopt -> re-prep) in world-2 land on
+    the same fiber positions as the single-process run — the distributed
+    step loop, not just one solve."""
+    with tempfile.TemporaryDirectory() as td:
+        init_file = os.path.join(td, "pg")
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        procs = [ctx.Process(target=_dist_multistep_worker,
+                             args=(r, WORLD, init_file, q))
+                 for r in range(WORLD)]
+        for p in procs:
+            p.start()
+        results = {}
+        for _ in range(WORLD):
+            rank, xs = q.get(timeout=250)
+            results[rank] = xs
+        for p in procs:
+            p.join(timeout=60)
+            assert p.exitcode == 0
+
+    from skellysim_amd.system_fd import SystemFD, Shell
+    from skellysim_amd.sharded import shard_range
+    from oracle_backend import OracleBackend
+    fx, fibers, bg = _make_problem()
+    shell = Shell(fx["nodes"], fx["normals"],
+                  fx["stresslet_plus_complementary"], fx["M_inv"])
+    sys_ = SystemFD(fibers, eta=1.0, dt=0.05, shell=shell,
+                    backend=OracleBackend(), background_flow=bg)
+    for _ in range(3):
+        info = sys_.step(tol=1e-11, maxiter=300, restart=100)
+        assert info["converged"]
+
+    dist_xs = []
+    for r in range(WORLD):
+        dist_xs.extend(results[r])
+    assert len(dist_xs) == len(fibers)
+    for xd, f in zip(dist_xs, sys_.fibers):
+        rel = np.linalg.norm(xd - f.x) / max(np.linalg.norm(f.x), 1e-30)
+        assert rel < 1e-6, rel
