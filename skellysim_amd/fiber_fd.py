@@ -153,9 +153,15 @@ class FiberFD:
 
     def __init__(self, x, length, bending_rigidity, eta, radius=0.0125,
                  force_scale=0.0, minus_clamped=False, penalty_param=500.0,
-                 beta_tstep=1.0):
-        self.x = np.ascontiguousarray(np.asarray(x, float).T
-                                      if np.asarray(x).shape[0] != 3 else x, dtype=float)
+                 beta_tstep=1.0, layout=None):
+        x = np.asarray(x, float)
+        if x.shape == (3, 3) and layout is None:
+            raise ValueError(
+                "ambiguous (3, 3) node array: pass layout='coords_major' "
+                "for (3, n) input or layout='nodes_major' for (n, 3)")
+        if layout == "nodes_major" or (layout is None and x.shape[0] != 3):
+            x = x.T
+        self.x = np.ascontiguousarray(x, dtype=float)
         if self.x.shape[0] != 3:
             raise ValueError("x must be (3, n) or (n, 3)")
         self.n_nodes = self.x.shape[1]

@@ -251,12 +251,23 @@ class DistributedSystemFD(SystemFD):
         res = np.zeros_like(x)
         x_fib = x[: self.fiber_sol_size]
         if self.fibers:
-            if self._fiber_lu_solve is None:
-                A_batch = np.stack([f.A for f in self.fibers])
-                self._fiber_lu_solve = self.backend.batched_lu(A_batch)
-            m = 4 * self.fibers[0].n_nodes
-            sol = self._fiber_lu_solve(x_fib.reshape(len(self.fibers), m))
-            res[: self.fiber_sol_size] = sol.reshape(-1)
+            if self._uniform:
+                if self._fiber_lu_solve is None:
+                    A_batch = np.stack([f.A for f in self.fibers])
+                    self._fiber_lu_solve = self.backend.batched_lu(A_batch)
+                m = 4 * self.fibers[0].n_nodes
+                sol = self._fiber_lu_solve(x_fib.reshape(len(self.fibers), m))
+                res[: self.fiber_sol_size] = sol.reshape(-1)
+            else:
+                # mixed-discretization local block: per-fiber LU solves
+                # (same fallback SystemFD.apply_preconditioner takes)
+                if self._fiber_lu_solve is None:
+                    self._fiber_lu_solve = [
+                        self.backend.batched_lu(f.A[None])
+                        for f in self.fibers]
+                for (f, a, b), solve in zip(self._fiber_slices(),
+                                            self._fiber_lu_solve):
+                    res[a:b] = solve(x_fib[a:b][None])[0]
         if self.shell:
             sh = slice(self.fiber_sol_size,
                        self.fiber_sol_size + self.shell_sol_size)
@@ -390,6 +401,14 @@ class DistributedSystemFD(SystemFD):
             for b in self.bodies:
                 b.step(self.dt, full[off: off + b.solution_size])
                 off += b.solution_size
+            # repin local attached fibers to the moved nucleation sites
+            # (f_c_fd.cpp:308-316 via system.cpp:488; bodies are replicated
+            # on every rank so each rank repins its own fiber block)
+            for f in self.fibers:
+                ib, js = getattr(f, "binding_site", (-1, -1))
+                if ib >= 0:
+                    delta = self.bodies[ib].nucleation_sites[js] - f.x[:, 0]
+                    f.x += delta[:, None]
         return info
 
 

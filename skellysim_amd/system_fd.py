@@ -763,13 +763,28 @@ class SystemFD:
         return info
 
     def step(self, tol=1e-10, maxiter=200, restart=None):
-        """system.cpp:482-493: solve then adopt positions."""
+        """system.cpp:482-493: solve then adopt positions, then repin
+        body-attached fibers to their (moved) nucleation sites
+        (FiberContainerFiniteDifference::repin_to_bodies,
+        fiber_container_finite_difference.cpp:308-316 via system.cpp:488)."""
         info = self.solve(tol=tol, maxiter=maxiter, restart=restart)
         for f, a, b in self._fiber_slices():
             f.step(self.solution[a:b])
         for b, a, bb in self._body_sol_slices():
             b.step(self.dt, self.solution[a:bb])
+        self.repin_to_bodies()
         return info
+
+    def repin_to_bodies(self):
+        """Translate every body-attached fiber so its minus end coincides
+        with the body's nucleation site again — fiber and body do not move
+        exactly together under finite dt
+        (fiber_container_finite_difference.cpp:308-316)."""
+        for f in self.fibers:
+            ib, js = getattr(f, "binding_site", (-1, -1))
+            if ib >= 0:
+                delta = self.bodies[ib].nucleation_sites[js] - f.x[:, 0]
+                f.x += delta[:, None]
 
     # ---- adaptive time-stepping driver (System::run, system.cpp:516-570) --
     def fiber_error(self):

@@ -133,17 +133,37 @@ class TrajectoryWriter:
 
 
 def _pack_rng(rng):
+    """Serialize as a list of [string, string] pairs — the same msgpack
+    shape as the reference's vector<pair<string,string>> rng_state
+    (io_maps.hpp:31-41), so the reference binary can at least parse (and
+    skip) the field; the VALUES are this engine's numpy PCG64 state
+    (128-bit ints as decimal strings), not Philox counters."""
     if rng is None:
         return []
     st = rng.bit_generator.state
-    return {"bit_generator": st["bit_generator"],
-            "state": {k: str(v) for k, v in st["state"].items()},
-            "has_uint32": int(st.get("has_uint32", 0)),
-            "uinteger": int(st.get("uinteger", 0))}
+    pairs = [["bit_generator", str(st["bit_generator"])]]
+    pairs += [["state." + k, str(v)] for k, v in st["state"].items()]
+    pairs += [["has_uint32", str(int(st.get("has_uint32", 0)))],
+              ["uinteger", str(int(st.get("uinteger", 0)))]]
+    return pairs
 
 
 def _restore_rng(rng, packed):
-    if rng is None or not isinstance(packed, dict):
+    if rng is None:
+        return
+    if isinstance(packed, (list, tuple)):  # current pair-list format
+        packed = {k: v for k, v in packed}
+        if packed.get("bit_generator") != \
+                rng.bit_generator.state["bit_generator"]:
+            return
+        rng.bit_generator.state = {
+            "bit_generator": packed["bit_generator"],
+            "state": {k[len("state."):]: int(v) for k, v in packed.items()
+                      if k.startswith("state.")},
+            "has_uint32": int(packed.get("has_uint32", 0)),
+            "uinteger": int(packed.get("uinteger", 0))}
+        return
+    if not isinstance(packed, dict):  # legacy map format
         return
     if packed.get("bit_generator") != rng.bit_generator.state["bit_generator"]:
         return

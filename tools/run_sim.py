@@ -90,9 +90,13 @@ def main():
                             {**cfg, "params": {**p, "periphery_interaction_flag": True}}),
                         tol=tol, on_accept=on_accept)
         wall = time.perf_counter() - t0
-    print(f"{len(hist)} accepted steps to t={hist[-1]['time']:.4f} in {wall:.1f}s "
-          f"({len(hist)/wall:.3f} steps/s); iters/step: "
-          f"{[h['iters'] for h in hist[:12]]}", flush=True)
+    if hist:
+        print(f"{len(hist)} accepted steps to t={hist[-1]['time']:.4f} in {wall:.1f}s "
+              f"({len(hist)/wall:.3f} steps/s); iters/step: "
+              f"{[h['iters'] for h in hist[:12]]}", flush=True)
+    else:
+        print(f"0 accepted steps (time already >= t_final) in {wall:.1f}s",
+              flush=True)
 
 
 if __name__ == "__main__":
