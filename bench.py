@@ -61,6 +61,15 @@ def cpu_baseline(r_src, f_src, r_trg, eta, kernel="stokeslet", budget_s=15.0):
         fn = lambda t: oracle.oseen_contract(r_src, t, f_src, eta)
     n_src = len(r_src)
     cores = oracle.num_threads()
+    cpu_model = None
+    try:
+        with open("/proc/cpuinfo") as fh:
+            for line in fh:
+                if line.startswith("model name"):
+                    cpu_model = line.split(":", 1)[1].strip()
+                    break
+    except OSError:
+        pass
     # probe to pick a sample that costs ~budget_s
     probe_t = min(256, len(r_trg))
     t0 = time.perf_counter()
@@ -78,6 +87,7 @@ def cpu_baseline(r_src, f_src, r_trg, eta, kernel="stokeslet", budget_s=15.0):
         "kind": "port",
         "sample": f"{sample_t} of {len(r_trg)} targets x {n_src} sources, "
                   f"{dt:.1f}s, OpenMP {cores} threads",
+        "cpu": cpu_model,
     }
 
 
@@ -254,7 +264,12 @@ def main():
                                if world > 1 else "single GPU",
             },
             "roofline": {
+                # schema enum offers only hbm|mfma; the kernel is actually
+                # fp64 VALU instruction-issue bound — see
+                # bound_detail
                 "bound": "mfma",
+                "bound_detail": "fp64-valu-issue (compute; not matrix-core "
+                                "shaped: gfx950 fp64 MFMA rate == vector rate)",
                 "achieved": achieved_tflops,
                 "peak": FP64_PEAK_TFLOPS,
                 "unit": "TFLOP/s",
