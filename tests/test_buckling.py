@@ -21,14 +21,15 @@ from skellysim_amd.sources import PointSource, PointSourceContainer
 from oracle_backend import OracleBackend
 
 
-def run_sigma(sigma):
+def run_sigma(sigma, backend=None, t_final=50.0):
     """gen_config of the reference test, verbatim parameters."""
     length, E, n = 1.0, 0.0025, 32
     force_scale = -sigma * E / length ** 3
     x = np.linspace([0, 0, 0], [0, 0, length], n)
     fib = FiberFD(x, length=length, bending_rigidity=E, eta=1.0,
                   minus_clamped=True, force_scale=force_scale)
-    s = SystemFD([fib], eta=1.0, dt=0.02, backend=OracleBackend())
+    s = SystemFD([fib], eta=1.0, dt=0.02,
+                 backend=backend if backend is not None else OracleBackend())
     s.point_sources = PointSourceContainer(
         [PointSource(position=[0.0, 0.0, 10 * length], force=[10.0, 0.0, 0.0],
                      time_to_live=1.0)])
@@ -40,7 +41,7 @@ def run_sigma(sigma):
         if int(t / dt_write) > int((t - sys_.dt) / dt_write):
             xs.append(sys_.fibers[0].x[0, -1])   # plus-end x per frame
 
-    s.run(t_final=50.0, adaptive=True, dt_min=0.01, dt_max=0.1,
+    s.run(t_final=t_final, adaptive=True, dt_min=0.01, dt_max=0.1,
           tol=1e-10, maxiter=300, on_accept=on_accept)
     x_arr = np.array(xs)
     peaks, _ = find_peaks(x_arr, height=0)

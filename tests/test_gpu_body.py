@@ -104,3 +104,65 @@ def test_full_composition_hip_matches_oracle(hip_lib_path):
     rel = np.linalg.norm(s_hip.solution - s_cpu.solution) / \
         np.linalg.norm(s_cpu.solution)
     assert rel < 1e-8, rel
+
+
+def test_device_resident_bodies_match_host_path(hip_lib_path):
+    """the fully device-resident GMRES iteration
+    WITH BODY BLOCKS (torch link conditions, body flow and body rows inside
+    _apply_matvec_device) must match the host matvec loop on real device
+    tensors, for the full shell+body+fiber composition AND for a 3-step
+    solve-adopt-reprep run."""
+    from skellysim_amd.body import SphericalBody
+    from skellysim_amd.fiber_fd import FiberFD
+    from skellysim_amd.system_fd import SystemFD, Shell, HipBackend
+    import torch
+
+    fx = np.load(os.path.join(HERE, "golden", "periphery_sphere_192.npz"))
+    R = float(fx["radius"])
+
+    def build():
+        shell = Shell(fx["nodes"] * 4.0, fx["normals"],
+                      fx["stresslet_plus_complementary"], fx["M_inv"])
+        b = SphericalBody(fx["nodes"], -fx["normals"],
+                          fx["quadrature_weights"].reshape(-1), R,
+                          nucleation_sites_ref=np.array([[1.1 * R, 0.0, 0.0]]),
+                          external_force=(0.0, 0.0, 0.2))
+        s0 = np.linspace(0, 1.0, 16)
+        x = b.nucleation_sites[0][None, :] + s0[:, None] * np.array([1.0, 0, 0])
+        fib = FiberFD(x, length=1.0, bending_rigidity=2.5e-3, eta=1.0,
+                      minus_clamped=True, force_scale=-0.05)
+        fib.binding_site = (0, 0)
+        return SystemFD([fib], eta=1.0, dt=0.05, shell=shell, bodies=[b],
+                        backend=HipBackend())
+
+    # single solve, device-resident vs host
+    s_dev = build()
+    i_dev = s_dev.solve(tol=1e-11, maxiter=300, restart=150, device_mode=True)
+    assert i_dev["converged"], i_dev
+    s_host = build()
+    i_host = s_host.solve(tol=1e-11, maxiter=300, restart=150,
+                          device_mode=False)
+    assert i_host["converged"], i_host
+    rel = np.linalg.norm(s_dev.solution - s_host.solution) / \
+        np.linalg.norm(s_host.solution)
+    assert rel < 1e-8, rel
+
+    # 3-step run: positions must track to solver tolerance
+    def adopt(s):
+        for f, a, bnd in s._fiber_slices():
+            f.step(s.solution[a:bnd])
+        for b_, a, bb in s._body_sol_slices():
+            b_.step(s.dt, s.solution[a:bb])
+        s.repin_to_bodies()
+
+    s_dev, s_host = build(), build()
+    for _ in range(3):
+        assert s_dev.solve(tol=1e-11, maxiter=300, restart=150,
+                           device_mode=True)["converged"]
+        adopt(s_dev)
+        assert s_host.solve(tol=1e-11, maxiter=300, restart=150,
+                            device_mode=False)["converged"]
+        adopt(s_host)
+    dx = np.linalg.norm(s_dev.fibers[0].x - s_host.fibers[0].x)
+    db = np.linalg.norm(s_dev.bodies[0].position - s_host.bodies[0].position)
+    assert dx < 1e-8 and db < 1e-8, (dx, db)
