@@ -147,14 +147,18 @@ class BatchedLU:
 
     @staticmethod
     def _use_trsm():
-        """SKELLY_LU_TRSM=1 routes the per-iteration solves through batched
+        """Default ON: the per-iteration solves go through batched
         solve_triangular (rocBLAS trsm) instead of torch.linalg.lu_solve
-        (magma batched trsv on ROCm). Round-2 experiment knob: magma's
-        batched calls build per-call device pointer arrays, a use-after-free
-        candidate for the deep-queue (sync-cadence) corruption documented
-        in gmres.py."""
+        (magma batched trsv on ROCm). The round-2 experiment matrix
+        (profiles/cadence_matrix_r02.md) isolated magma's lu_solve as the
+        source of the deep-queue corruption that poisoned large solves at
+        GMRES sync cadence 8: with trsm the cadence-8 solve is BITWISE
+        equal to cadence 1 at config-5 scale, with magma the second solve
+        of a process deterministically diverges (magma's batched calls
+        build per-call device pointer arrays — a use-after-free under deep
+        stream queues). SKELLY_LU_TRSM=0 restores magma for comparison."""
         import os
-        return os.environ.get("SKELLY_LU_TRSM", "0") == "1"
+        return os.environ.get("SKELLY_LU_TRSM", "1") == "1"
 
     def _perm_from_pivots(self):
         if getattr(self, "_perm", None) is None:

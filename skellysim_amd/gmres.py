@@ -54,19 +54,17 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
     device stream while the host processes earlier Hessenberg columns.
     The host bookkeeping is identical for any cadence (deferred, not
     changed), and CPU tensors are bitwise-identical across cadences
-    (tests/test_gmres.py). DEFAULT 1: on MI355X, cadence 8 deterministically
-    corrupts large solves (config-5 scale, ~500k unknowns, >70 iterations)
-    in their SECOND solve of a process — implicit residual diverges from the
-    true residual by ~5e-3 — while every pair kernel checks out bitwise in
-    isolation (tools/debug_matvec.py) and the same run is clean at cadence
-    1; root cause (suspected async hazard in a library call that the
-    per-iteration sync masks) not yet isolated. Opt in to >1 via the
-    SKELLY_GMRES_SYNC_CADENCE env var; ~20% faster per iteration when it
-    holds.
+    (tests/test_gmres.py). DEFAULT 8 (~20% faster per iteration at
+    config-5 scale): safe since the round-2 experiment matrix
+    (profiles/cadence_matrix_r02.md) isolated the deep-queue corruption
+    that cadence>1 used to trigger to magma's torch.linalg.lu_solve in the
+    preconditioner — with the trsm-based LU solves (batched.py, now the
+    default) cadence 8 is bitwise equal to cadence 1 on device at config-5
+    scale. Override via the SKELLY_GMRES_SYNC_CADENCE env var.
     """
     if sync_cadence is None:
         import os
-        sync_cadence = int(os.environ.get("SKELLY_GMRES_SYNC_CADENCE", "1"))
+        sync_cadence = int(os.environ.get("SKELLY_GMRES_SYNC_CADENCE", "8"))
     if precond is None:
         precond = lambda v: v
     reduce_ = _make_reduce(distributed, group)

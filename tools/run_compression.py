@@ -48,7 +48,14 @@ def main():
     print(f"fiber plus end {frames[-1]['fiber_x'][:, -1]}")
     if args.t_final == 5.0:
         err = final_position_error(frames)
-        print(f"final-position total error vs reference pins: {err:.3e}")
+        print(f"final-position total error vs reference pins: {err:.3e} "
+              f"(reference gate 1e-5)")
+        from compression_common import velocity_field_error
+        for cand in (96, 97, 98):
+            if cand < len(frames):
+                ve, npts = velocity_field_error(frames, backend, frame_no=cand)
+                print(f"velocity-field error vs pins at frame {cand} "
+                      f"(t={frames[cand]['time']:.3f}, {npts} pts): {ve:.3e}")
     else:
         print("(partial run; pin comparison skipped)")
 
