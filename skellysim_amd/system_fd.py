@@ -535,11 +535,16 @@ class SystemFD:
         if self.bodies:
             # per-body blocks + fiber<->body link statics (body.py)
             blocks = []
+            from .batched import BatchedLU
             for b in self.bodies:
-                A_lu = torch.linalg.lu_factor(T(b._A_dense))
+                # BatchedLU(batch of 1) so the per-iteration solves take the
+                # trsm path — magma's lu_solve corrupts under deep stream
+                # queues (profiles/cadence_matrix_r02.md) and the device
+                # preconditioner runs at sync cadence 8 by default
                 blocks.append(dict(n=b.n_nodes,
                                    e=T(np.stack(b.e_sub)),     # (3, n, 3)
-                                   w=T(b.weights), K=T(b.K), lu=A_lu))
+                                   w=T(b.weights), K=T(b.K),
+                                   lu=BatchedLU(T(b._A_dense)[None])))
             d["bodies"] = blocks
             d["body_nodes"] = T(self.body_nodes())
             d["body_normals"] = T(self.body_normals())
@@ -721,9 +726,8 @@ class SystemFD:
             off = 4 * nf * n + sh_size
             for bd in d["bodies"]:
                 m = 3 * bd["n"] + 6
-                res[off: off + m] = torch.linalg.lu_solve(
-                    bd["lu"][0], bd["lu"][1],
-                    x[off: off + m].unsqueeze(-1)).squeeze(-1)
+                res[off: off + m] = bd["lu"].solve(
+                    x[off: off + m].unsqueeze(0)).squeeze(0)
                 off += m
         return res
 
