@@ -122,6 +122,24 @@ class BatchedLU:
         if mats.dtype != torch.float64:
             raise TypeError("BatchedLU expects fp64")
         self._host_lus = None
+        self._Ainv = None
+        if mats.is_cuda and self._mode() == "inv":
+            # explicit batched inverse, applied per iteration as ONE
+            # rocBLAS batched GEMM (torch.bmm) — no magma and no
+            # triangular-solve kernels inside the GMRES iteration at all.
+            # Safe for (right-)preconditioning: M^-1 is a FIXED linear
+            # operator whatever its rounding, GMRES converges on the true
+            # residual regardless; the factor-time inversion happens once
+            # per timestep under a shallow queue (prep syncs), where the
+            # deep-queue magma hazard (profiles/cadence_matrix_r02.md)
+            # does not apply. Probe-verified like every inverse here.
+            try:
+                self._Ainv = torch.linalg.inv(mats)
+            except RuntimeError:
+                self._Ainv = None
+            if self._Ainv is not None and self._probe_ok(mats):
+                return
+            self._Ainv = None  # probe failed: fall through to LU paths
         self.LU, self.pivots = _lu_factor(mats)
         if mats.is_cuda and not self._probe_ok(mats):
             # same defensive posture as robust_inv: a silently corrupt
@@ -146,19 +164,35 @@ class BatchedLU:
         return math.isfinite(rel) and rel < rtol
 
     @staticmethod
-    def _use_trsm():
-        """Default ON: the per-iteration solves go through batched
-        solve_triangular (rocBLAS trsm) instead of torch.linalg.lu_solve
-        (magma batched trsv on ROCm). The round-2 experiment matrix
-        (profiles/cadence_matrix_r02.md) isolated magma's lu_solve as the
-        source of the deep-queue corruption that poisoned large solves at
-        GMRES sync cadence 8: with trsm the cadence-8 solve is BITWISE
-        equal to cadence 1 at config-5 scale, with magma the second solve
-        of a process deterministically diverges (magma's batched calls
-        build per-call device pointer arrays — a use-after-free under deep
-        stream queues). SKELLY_LU_TRSM=0 restores magma for comparison."""
+    def _mode():
+        """Per-iteration solve strategy (SKELLY_LU_MODE):
+          "inv"   (default) — explicit batched inverse applied with one
+                  torch.bmm per apply; fastest and keeps magma AND
+                  triangular kernels out of the GMRES iteration entirely.
+          "trsm"  — batched solve_triangular (correct under deep queues,
+                  measured bitwise-stable at cadence 8).
+          "magma" — torch.linalg.lu_solve: CORRUPTS under deep stream
+                  queues (sync cadence > 1) — magma's batched trsv builds
+                  per-call device pointer arrays, a use-after-free that
+                  the round-2 experiment matrix isolated
+                  (profiles/cadence_matrix_r02.md). Keep only for
+                  retesting future ROCm/magma.
+        The legacy SKELLY_LU_TRSM=1/0 env maps to trsm/magma when
+        SKELLY_LU_MODE is unset."""
         import os
-        return os.environ.get("SKELLY_LU_TRSM", "1") == "1"
+        mode = os.environ.get("SKELLY_LU_MODE")
+        if mode in ("inv", "trsm", "magma"):
+            return mode
+        legacy = os.environ.get("SKELLY_LU_TRSM")
+        if legacy == "1":
+            return "trsm"
+        if legacy == "0":
+            return "magma"
+        return "inv"
+
+    @classmethod
+    def _use_trsm(cls):
+        return cls._mode() != "magma"
 
     def _perm_from_pivots(self):
         if getattr(self, "_perm", None) is None:
@@ -174,6 +208,10 @@ class BatchedLU:
 
     def solve(self, rhs):
         """rhs: (n_fibers, m) or (n_fibers, m, k) -> same shape solution."""
+        if self._Ainv is not None:
+            if rhs.dim() == 2:
+                return torch.bmm(self._Ainv, rhs.unsqueeze(-1)).squeeze(-1)
+            return torch.bmm(self._Ainv, rhs)
         if self._host_lus is not None:
             import numpy as _np
             import scipy.linalg as scla
