@@ -734,10 +734,11 @@ class SystemFD:
     def solve(self, tol=1e-10, maxiter=200, restart=None, device_mode=None):
         """system.cpp:464-478 via the engine GMRES (right-preconditioned,
         ICGS — solver_hydro.cpp:64-87). With uniform fibers on the HIP
-        backend the whole iteration runs device-resident. device_mode=None
-        auto-selects (host path when bodies are present — their device
-        blocks exist but are validated on CPU tensors only this round;
-        pass device_mode=True to opt in)."""
+        backend the whole iteration runs device-resident, INCLUDING body
+        blocks (torch link conditions + body flow + body rows — default
+        since the round-2 GPU validation, tests/test_gpu_body.py::
+        test_device_resident_bodies_match_host_path). device_mode=False
+        forces the host matvec loop."""
         import torch
         from .gmres import gmres
 
@@ -746,7 +747,7 @@ class SystemFD:
             restart = min(200, maxiter)
 
         if device_mode is None:
-            device_mode = (self._uniform and self.fibers and not self.bodies
+            device_mode = (bool(self.fibers) and self._uniform
                            and isinstance(self.backend, HipBackend))
         if device_mode:
             self._build_device_operators()
