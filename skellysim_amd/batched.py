@@ -133,10 +133,21 @@ class BatchedLU:
             # per timestep under a shallow queue (prep syncs), where the
             # deep-queue magma hazard (profiles/cadence_matrix_r02.md)
             # does not apply. Probe-verified like every inverse here.
+            # force the default (hipSOLVER) backend for the batched
+            # inversion: once robust_inv's 18000^2 probe failure has latched
+            # magma globally, magma's getri LOOPS over the 4000-matrix batch
+            # (~16 s/step measured at config 5 vs 8 ms batched —
+            # profiles/cadence_matrix_r02.md addendum)
+            global _magma_latched
+            if _magma_latched:
+                torch.backends.cuda.preferred_linalg_library("default")
             try:
                 self._Ainv = torch.linalg.inv(mats)
             except RuntimeError:
                 self._Ainv = None
+            finally:
+                if _magma_latched:
+                    torch.backends.cuda.preferred_linalg_library("magma")
             if self._Ainv is not None and self._probe_ok(mats):
                 return
             self._Ainv = None  # probe failed: fall through to LU paths

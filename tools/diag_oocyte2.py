@@ -108,6 +108,19 @@ def main():
         if "long" in cases:
             one_step("oocyte-long", place_fibers(fx, own["normals"], 4000, 32),
                      shell, maxiter=2000, restart=500)
+        if "refenv" in cases:
+            # the reference's EXACT solver envelope (Belos defaults:
+            # restart=Num Blocks=300, maxiter=1000, tol from config=1e-8)
+            # on the reference example's fiber protocol (3000 fibers,
+            # >= 0.1 separation, dt=1e-2)
+            one_step("oocyte-refenv",
+                     place_fibers(fx, own["normals"], 3000, 32, ds_min=0.1),
+                     shell, maxiter=1000, restart=300)
+        if "longrestart" in cases:
+            # restart stagnation test: no restart inside the budget
+            one_step("oocyte-r1500",
+                     place_fibers(fx, own["normals"], 4000, 32),
+                     shell, maxiter=1500, restart=1500)
 
 
 if __name__ == "__main__":
