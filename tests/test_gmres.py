@@ -154,3 +154,40 @@ def test_distributed_gmres_gloo():
     b = rng.uniform(-1, 1, n)
     x = np.concatenate([res[0], res[1]])
     assert np.linalg.norm(A @ x - b) / np.linalg.norm(b) < 1e-9
+
+
+def test_warm_start_systemfd_matches_cold():
+    """SystemFD.solve(warm_start=True) seeds GMRES with the previous
+    solution: the converged answer matches the cold-start solve to solver
+    tolerance and takes fewer iterations on the second step."""
+    import numpy as np
+    from skellysim_amd.fiber_fd import FiberFD
+    from skellysim_amd.system_fd import SystemFD
+    from oracle_backend import OracleBackend
+
+    def build():
+        x = np.linspace([0, 0, 0], [0, 0, 1.0], 24)
+        f = FiberFD(x, length=1.0, bending_rigidity=2.5e-3, eta=1.0,
+                    minus_clamped=True, force_scale=-0.08)
+        return SystemFD([f], eta=1.0, dt=0.05, backend=OracleBackend())
+
+    s_cold, s_warm = build(), build()
+    # step 1: identical (warm has no previous solution yet)
+    i1c = s_cold.step(tol=1e-11, maxiter=200, restart=100)
+    s_warm.solve(tol=1e-11, maxiter=200, restart=100, warm_start=True)
+    for f, a, b in s_warm._fiber_slices():
+        f.step(s_warm.solution[a:b])
+    s_warm.repin_to_bodies()
+    assert np.array_equal(s_cold.solution, s_warm.solution)
+
+    # step 2: warm start converges to the same solution in fewer iterations
+    i2c = s_cold.solve(tol=1e-11, maxiter=200, restart=100)
+    i2w = s_warm.solve(tol=1e-11, maxiter=200, restart=100, warm_start=True)
+    assert i2c["converged"] and i2w["converged"]
+    assert i2w["iters"] <= i2c["iters"], (i2w["iters"], i2c["iters"])
+    # the warm start's initial residual is far below the cold start's
+    assert i2w["residuals"][0] < 1e-2 * i2c["residuals"][0], \
+        (i2w["residuals"][0], i2c["residuals"][0])
+    rel = np.linalg.norm(s_warm.solution - s_cold.solution) / \
+        np.linalg.norm(s_cold.solution)
+    assert rel < 1e-8, rel
