@@ -94,7 +94,7 @@ def main():
         del shell
         torch.cuda.empty_cache()
 
-    if "count" in cases or "long" in cases:
+    if {"count", "long", "refenv", "longrestart"} & set(cases):
         fx = np.load(os.path.join(repo, "tests", "golden", "oocyte_nodes.npz"))
         own = surface_of_revolution_normals_weights(
             fx["nodes"], float(fx["envelope_T"]), float(fx["envelope_p1"]),
@@ -121,6 +121,20 @@ def main():
             one_step("oocyte-r1500",
                      place_fibers(fx, own["normals"], 4000, 32),
                      shell, maxiter=1500, restart=1500)
+        if "protocol" in cases:
+            # multi-step reference-envelope protocol with WARM STARTS:
+            # after step 1 each solve corrects only the O(dt) change
+            os.environ["SKELLY_WARM_START"] = "1"
+            s = SystemFD(place_fibers(fx, own["normals"], 3000, 32,
+                                      ds_min=0.1),
+                         eta=1.0, dt=0.01, shell=shell, backend=HipBackend())
+            for k in range(5):
+                t0 = time.perf_counter()
+                info = s.step(tol=1e-8, maxiter=1000, restart=300)
+                print(f"[oocyte-protocol-warm] step {k}: "
+                      f"iters={info['iters']} conv={info['converged']} "
+                      f"({time.perf_counter()-t0:.1f}s)", flush=True)
+            os.environ.pop("SKELLY_WARM_START")
 
 
 if __name__ == "__main__":
