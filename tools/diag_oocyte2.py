@@ -122,15 +122,15 @@ def main():
                      place_fibers(fx, own["normals"], 4000, 32),
                      shell, maxiter=1500, restart=1500)
         if "protocol" in cases:
-            # multi-step reference-envelope protocol with WARM STARTS:
-            # after step 1 each solve corrects only the O(dt) change
+            # multi-step protocol with WARM STARTS and NO in-budget restart
+            # (restart stagnation kills the dense-packing solve: restart
+            # 300 stalls forever, restart >= ~800 converges in ~732 iters)
             os.environ["SKELLY_WARM_START"] = "1"
-            s = SystemFD(place_fibers(fx, own["normals"], 3000, 32,
-                                      ds_min=0.1),
+            s = SystemFD(place_fibers(fx, own["normals"], 4000, 32),
                          eta=1.0, dt=0.01, shell=shell, backend=HipBackend())
             for k in range(5):
                 t0 = time.perf_counter()
-                info = s.step(tol=1e-8, maxiter=1000, restart=300)
+                info = s.step(tol=1e-8, maxiter=1500, restart=1500)
                 print(f"[oocyte-protocol-warm] step {k}: "
                       f"iters={info['iters']} conv={info['converged']} "
                       f"({time.perf_counter()-t0:.1f}s)", flush=True)

@@ -32,7 +32,15 @@ def main():
     ap.add_argument("--geometry", default="sphere", choices=["sphere", "oocyte"],
                     help="oocyte = the reference example's surface of revolution "
                          "(tests/golden/oocyte_nodes.npz)")
+    ap.add_argument("--tol", type=float, default=1e-10)
+    ap.add_argument("--maxiter", type=int, default=300)
+    ap.add_argument("--restart", type=int, default=150)
+    ap.add_argument("--warm", action="store_true",
+                    help="warm-start each solve from the previous solution")
+    ap.add_argument("--dt", type=float, default=0.025)
     args = ap.parse_args()
+    if args.warm:
+        os.environ["SKELLY_WARM_START"] = "1"
 
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     geom = "sphere_6000_nodes.npz" if args.geometry == "sphere" else "oocyte_nodes.npz"
@@ -92,13 +100,14 @@ def main():
                               minus_clamped=True, force_scale=-0.05))
     print(f"placed {len(fibers)} fibers")
 
-    sys_ = SystemFD(fibers, eta=1.0, dt=0.025, shell=shell, backend=HipBackend())
+    sys_ = SystemFD(fibers, eta=1.0, dt=args.dt, shell=shell, backend=HipBackend())
     print(f"solution size: {sys_.fiber_sol_size + sys_.shell_sol_size}")
 
     t0 = time.perf_counter()
     iters = []
     for k in range(args.steps):
-        info = sys_.step(tol=1e-10, maxiter=300, restart=150)
+        info = sys_.step(tol=args.tol, maxiter=args.maxiter,
+                         restart=args.restart)
         iters.append(info["iters"])
         assert info["converged"], info
     dt_total = time.perf_counter() - t0
