@@ -94,7 +94,8 @@ def main():
         del shell
         torch.cuda.empty_cache()
 
-    if {"count", "long", "refenv", "longrestart"} & set(cases):
+    oocyte_cases = {"count", "long", "refenv", "longrestart", "protocol"}
+    if oocyte_cases & set(cases):
         fx = np.load(os.path.join(repo, "tests", "golden", "oocyte_nodes.npz"))
         own = surface_of_revolution_normals_weights(
             fx["nodes"], float(fx["envelope_T"]), float(fx["envelope_p1"]),
