@@ -94,7 +94,8 @@ def main():
         del shell
         torch.cuda.empty_cache()
 
-    oocyte_cases = {"count", "long", "refenv", "longrestart", "protocol"}
+    oocyte_cases = {"count", "long", "refenv", "longrestart", "protocol",
+                    "refproto"}
     if oocyte_cases & set(cases):
         fx = np.load(os.path.join(repo, "tests", "golden", "oocyte_nodes.npz"))
         own = surface_of_revolution_normals_weights(
@@ -136,6 +137,22 @@ def main():
                       f"iters={info['iters']} conv={info['converged']} "
                       f"({time.perf_counter()-t0:.1f}s)", flush=True)
             os.environ.pop("SKELLY_WARM_START")
+        if "refproto" in cases:
+            # the reference example's OWN protocol: 3000 fibers at >= 0.1
+            # separation (move_fibers_to_surface ds_min), dt 1e-2, the
+            # reference ADAPTIVE loop (rejection + dt shrink on
+            # non-convergence), engine solver envelope (full restart)
+            s = SystemFD(place_fibers(fx, own["normals"], 3000, 32,
+                                      ds_min=0.1),
+                         eta=1.0, dt=0.01, shell=shell, backend=HipBackend())
+            t0 = time.perf_counter()
+            hist = s.run(t_final=0.05, adaptive=True, dt_min=1e-4,
+                         dt_max=0.01, tol=1e-8, maxiter=1500, restart=1500)
+            wall = time.perf_counter() - t0
+            print(f"[oocyte-refproto] {len(hist)} accepted steps in "
+                  f"{wall:.1f}s ({len(hist)/wall:.3f} steps/s); "
+                  f"iters: {[h['iters'] for h in hist]}; "
+                  f"dt: {[round(h['dt'], 5) for h in hist]}", flush=True)
 
 
 if __name__ == "__main__":
