@@ -95,7 +95,7 @@ def main():
         torch.cuda.empty_cache()
 
     oocyte_cases = {"count", "long", "refenv", "longrestart", "protocol",
-                    "refproto"}
+                    "refproto", "sep1500"}
     if oocyte_cases & set(cases):
         fx = np.load(os.path.join(repo, "tests", "golden", "oocyte_nodes.npz"))
         own = surface_of_revolution_normals_weights(
@@ -110,6 +110,12 @@ def main():
         if "long" in cases:
             one_step("oocyte-long", place_fibers(fx, own["normals"], 4000, 32),
                      shell, maxiter=2000, restart=500)
+        if "sep1500" in cases:
+            # why does the reference-style 3000@0.1 placement fail where
+            # 4000-at-nodes converges at full restart? residual trace:
+            one_step("oocyte-3000@0.1-r1500",
+                     place_fibers(fx, own["normals"], 3000, 32, ds_min=0.1),
+                     shell, maxiter=1500, restart=1500)
         if "refenv" in cases:
             # the reference's EXACT solver envelope (Belos defaults:
             # restart=Num Blocks=300, maxiter=1000, tol from config=1e-8)

@@ -75,6 +75,20 @@ def profile(label, s, restart=150):
 
     mv = t_ms(lambda: s._apply_matvec_device(x))
     pc = t_ms(lambda: s._apply_precond_device(x))
+    # preconditioner sub-parts
+    d = s._dev
+    nf, n = d["nf"], d["n"]
+    x_fib = x[: 4 * nf * n].reshape(nf, 4 * n).contiguous()
+    fib_ms = t_ms(lambda: d["lu"].solve(x_fib))
+    if s.shell:
+        sh = x[4 * nf * n: 4 * nf * n + s.shell_sol_size].contiguous()
+        shell_ms = t_ms(lambda: torch.mv(d["sh_Minv"], sh))
+        shell_t_ms = t_ms(lambda: torch.mv(d["sh_Minv"].t(), sh))
+    else:
+        shell_ms = shell_t_ms = 0.0
+    print(f"  precond parts: fiber-solve {fib_ms:.2f} ms, "
+          f"shell M_inv mv {shell_ms:.2f} ms (transposed-layout mv "
+          f"{shell_t_ms:.2f} ms)", flush=True)
 
     t0 = time.perf_counter()
     info = s.solve(tol=1e-10, maxiter=300, restart=restart)
