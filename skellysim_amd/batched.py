@@ -148,10 +148,36 @@ class BatchedLU:
             finally:
                 if _magma_latched:
                     torch.backends.cuda.preferred_linalg_library("magma")
-            if self._Ainv is not None and self._probe_ok(mats):
+            # probe at 1e-4: the probe exists to catch CORRUPTION (O(1)
+            # garbage from the hipBLAS/hipSOLVER failure modes), not the
+            # cond*eps forward error of a legitimate explicit inverse —
+            # fiber operators at 4n=256 carry cond ~1e10 and a correct
+            # inverse legitimately probes ~1e-6..1e-5 (which as a fixed
+            # right-preconditioner costs nothing: GMRES converges on the
+            # true residual regardless)
+            if self._Ainv is not None and self._probe_ok(mats, rtol=1e-4):
                 return
-            self._Ainv = None  # probe failed: fall through to LU paths
+            self._Ainv = None  # corrupt or failed: fall through to LU
         self.LU, self.pivots = _lu_factor(mats)
+        if mats.is_cuda and self._mode() == "inv":
+            # inverse-from-LU: one shallow-queue magma lu_solve against I
+            # per timestep (safe — the deep-queue hazard is per-ITERATION
+            # calls), then bmm per iteration like the direct-inv path
+            m = mats.shape[-1]
+            eye = torch.eye(m, dtype=mats.dtype, device=mats.device) \
+                .expand_as(mats).contiguous()
+            try:
+                self._Ainv = torch.linalg.lu_solve(self.LU, self.pivots, eye)
+                torch.cuda.synchronize()
+            except RuntimeError:
+                self._Ainv = None
+            if self._Ainv is not None and self._probe_ok(mats, rtol=1e-4):
+                return
+            self._Ainv = None
+            import sys
+            print("BatchedLU: explicit-inverse probes failed (direct and "
+                  "from-LU); falling back to per-iteration triangular "
+                  "solves", file=sys.stderr, flush=True)
         if mats.is_cuda and not self._probe_ok(mats):
             # same defensive posture as robust_inv: a silently corrupt
             # device factorization must not poison solves

@@ -529,6 +529,16 @@ class SystemFD:
                 else T(self.shell.A)
             d["sh_Minv"] = self.shell.M_inv if torch.is_tensor(self.shell.M_inv) \
                 else T(self.shell.M_inv)
+            # transposed copies: rocBLAS dgemv's reduction (trans) path is
+            # ~1.5x the axpy path on these square operators (measured
+            # 0.96 vs 1.43 ms at 24576^2, tools/prof_iter.py), so store
+            # X^T contiguous and apply as mv(X_T.t(), v). HBM cost is one
+            # extra copy of each operator — trivial in 288 GB.
+            if getattr(self.shell, "_A_T", None) is None:
+                self.shell._A_T = d["sh_A"].t().contiguous()
+                self.shell._Minv_T = d["sh_Minv"].t().contiguous()
+            d["sh_A_T"] = self.shell._A_T
+            d["sh_Minv_T"] = self.shell._Minv_T
         from .batched import BatchedLU
         d["lu"] = BatchedLU(d["A"])
 
@@ -665,7 +675,7 @@ class SystemFD:
         if self.shell:
             v_shell = v_all[nf_nodes: nf_nodes + sh_nodes_n].reshape(-1)
             res[4 * nf_nodes: 4 * nf_nodes + sh_size] = \
-                d["sh_A"] @ x_shell + v_shell
+                torch.addmv(v_shell, d["sh_A_T"].t(), x_shell)
         if self.bodies:
             off = 0
             off_node = nf_nodes + sh_nodes_n
@@ -720,8 +730,8 @@ class SystemFD:
         res = torch.empty_like(x)
         res[: 4 * nf * n] = d["lu"].solve(x[: 4 * nf * n].reshape(nf, 4 * n)).reshape(-1)
         if self.shell:
-            res[4 * nf * n: 4 * nf * n + sh_size] = \
-                d["sh_Minv"] @ x[4 * nf * n: 4 * nf * n + sh_size]
+            res[4 * nf * n: 4 * nf * n + sh_size] = torch.mv(
+                d["sh_Minv_T"].t(), x[4 * nf * n: 4 * nf * n + sh_size])
         if self.bodies:
             off = 4 * nf * n + sh_size
             for bd in d["bodies"]:

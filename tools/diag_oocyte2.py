@@ -152,13 +152,34 @@ def main():
                                       ds_min=0.1),
                          eta=1.0, dt=0.01, shell=shell, backend=HipBackend())
             t0 = time.perf_counter()
-            hist = s.run(t_final=0.05, adaptive=True, dt_min=1e-4,
-                         dt_max=0.01, tol=1e-8, maxiter=1500, restart=1500)
+            accepted = 0
+            attempt = 0
+            while s.time < 0.05 and attempt < 20:
+                attempt += 1
+                s.backup()
+                ta = time.perf_counter()
+                info = s.step(tol=1e-8, maxiter=1500, restart=1500)
+                err = s.fiber_error()
+                ok = info["converged"] and err <= 0.1
+                print(f"[oocyte-refproto] attempt {attempt}: dt={s.dt:.5f} "
+                      f"iters={info['iters']} conv={info['converged']} "
+                      f"fiber_err={err:.3e} -> "
+                      f"{'ACCEPT' if ok else 'REJECT'} "
+                      f"({time.perf_counter()-ta:.1f}s)", flush=True)
+                if ok:
+                    accepted += 1
+                    s.time += s.dt
+                    if err <= 0.09:
+                        s.dt = min(0.01, s.dt * 1.2)
+                else:
+                    s.dt *= 0.5
+                    s.restore()
+                    if s.dt < 1e-4:
+                        print("[oocyte-refproto] dt underflow", flush=True)
+                        break
             wall = time.perf_counter() - t0
-            print(f"[oocyte-refproto] {len(hist)} accepted steps in "
-                  f"{wall:.1f}s ({len(hist)/wall:.3f} steps/s); "
-                  f"iters: {[h['iters'] for h in hist]}; "
-                  f"dt: {[round(h['dt'], 5) for h in hist]}", flush=True)
+            print(f"[oocyte-refproto] {accepted} accepted / {attempt} "
+                  f"attempts in {wall:.1f}s", flush=True)
 
 
 if __name__ == "__main__":
