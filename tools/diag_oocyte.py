@@ -40,11 +40,21 @@ def envelope_inside(fix, pts, margin=0.02):
 
 
 def place_fibers(fix, normals, n_fibers, n_nodes, ds_min=0.0, length=1.0,
-                 E=2.5e-3, seed=0):
+                 E=2.5e-3, seed=0, clearance=0.0):
     """Fibers seeded at shell nodes pointing inward; ds_min > 0 applies the
-    reference's move_fibers_to_surface minimum separation greedily."""
+    reference's move_fibers_to_surface minimum separation (minus ends
+    only). clearance > 0 additionally requires the WHOLE candidate fiber
+    to stay that far from every accepted fiber's nodes — inward fibers
+    from a concave surface FOCUS toward each other, and a near-contact
+    pair (separation below the node spacing) mis-integrates the stokeslet
+    quadrature into huge spurious velocities that reject timesteps at any
+    dt (the reference's placement checks minus ends only,
+    skelly_config.py:692-697, and suffers the same focusing)."""
+    from scipy.spatial import cKDTree
     fibers = []
     placed = []
+    cloud = None
+    cloud_n = 0
     order = np.random.default_rng(seed).permutation(len(fix["nodes"]))
     for i in order:
         if len(fibers) == n_fibers:
@@ -59,6 +69,20 @@ def place_fibers(fix, normals, n_fibers, n_nodes, ds_min=0.0, length=1.0,
         x = p[None, :] + s[:, None] * n[None, :]
         if not envelope_inside(fix, x):
             continue
+        if clearance > 0 and fibers:
+            if cloud is None or len(fibers) - cloud_n >= 64:
+                cloud = cKDTree(np.concatenate([f.x.T for f in fibers]))
+                cloud_n = len(fibers)
+            d, _ = cloud.query(x, k=1)
+            if np.min(d) < clearance:
+                continue
+            recent = [f.x.T for f in fibers[cloud_n:]]
+            if recent:
+                dr = np.linalg.norm(
+                    np.concatenate(recent)[None, :, :] - x[:, None, :],
+                    axis=2)
+                if dr.min() < clearance:
+                    continue
         placed.append(p)
         fibers.append(FiberFD(x, length=length, bending_rigidity=E, eta=1.0,
                               minus_clamped=True, force_scale=-0.05))

@@ -1,0 +1,129 @@
+#!/usr/bin/env python3
+"""BASELINE config 5 MEASURED protocol on the real oocyte geometry:
+~4k fibers (clearance-checked placement) inside the reference example's
+surface of revolution, reference adaptive loop (fiber-error acceptance +
+dt adaptation), warm-started full-restart GMRES, N accepted timesteps,
+then velocity-field parity vs the CPU oracle on the final state.
+
+Solver envelope per profiles/oocyte_r02.md: restart >= the slow-cluster
+size (~700 Krylov vectors at this packing) — the reference's Belos
+restart-300 envelope restart-stagnates here."""
+
+import argparse
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import numpy as np
+import torch
+
+from skellysim_amd.system_fd import SystemFD, HipBackend, Shell
+from skellysim_amd.periphery_precompute import assemble_shell_operator
+from skellysim_amd.precompute import surface_of_revolution_normals_weights
+from skellysim_amd.flows import velocity_at_targets
+from diag_oocyte import place_fibers
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--fibers", type=int, default=4000)
+    ap.add_argument("--steps", type=int, default=30,
+                    help="accepted timesteps to run")
+    ap.add_argument("--clearance", type=float, default=0.08)
+    ap.add_argument("--dt", type=float, default=0.01)
+    ap.add_argument("--tol", type=float, default=1e-8)
+    ap.add_argument("--maxiter", type=int, default=1500)
+    args = ap.parse_args()
+    os.environ["SKELLY_WARM_START"] = "1"
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    fx = np.load(os.path.join(repo, "tests", "golden", "oocyte_nodes.npz"))
+    dev = torch.device("cuda:0")
+    own = surface_of_revolution_normals_weights(
+        fx["nodes"], float(fx["envelope_T"]), float(fx["envelope_p1"]),
+        float(fx["envelope_p2"]), float(fx["envelope_length"]),
+        scale_factor=float(fx["scale_factor"]))
+
+    t0 = time.perf_counter()
+    A, M_inv = assemble_shell_operator(
+        torch.from_numpy(fx["nodes"]).to(dev),
+        torch.from_numpy(np.ascontiguousarray(own["normals"])).to(dev),
+        torch.from_numpy(np.ascontiguousarray(own["weights"])).to(dev))
+    torch.cuda.synchronize()
+    print(f"shell operator ({len(fx['nodes'])} nodes): "
+          f"{time.perf_counter()-t0:.1f}s", flush=True)
+    shell = Shell(fx["nodes"], own["normals"], A, M_inv)
+
+    fibers = place_fibers(fx, own["normals"], args.fibers, 32,
+                          clearance=args.clearance)
+    print(f"placed {len(fibers)} fibers (clearance {args.clearance})",
+          flush=True)
+    s = SystemFD(fibers, eta=1.0, dt=args.dt, shell=shell,
+                 backend=HipBackend())
+    print(f"solution size: {s.fiber_sol_size + s.shell_sol_size}", flush=True)
+
+    accepted = 0
+    attempts = 0
+    iters_hist, dt_hist = [], []
+    t0 = time.perf_counter()
+    while accepted < args.steps and attempts < 4 * args.steps:
+        attempts += 1
+        s.backup()
+        info = s.step(tol=args.tol, maxiter=args.maxiter,
+                      restart=args.maxiter)
+        err = s.fiber_error()
+        ok = info["converged"] and err <= 0.1
+        if ok:
+            accepted += 1
+            s.time += s.dt
+            iters_hist.append(info["iters"])
+            dt_hist.append(s.dt)
+            if err <= 0.09:
+                s.dt = min(args.dt, s.dt * 1.2)
+        else:
+            s.dt *= 0.5
+            s.restore()
+            if s.dt < 1e-5:
+                print("dt underflow — stopping", flush=True)
+                break
+        if attempts % 5 == 0:
+            el = time.perf_counter() - t0
+            print(f"  .. {accepted}/{attempts} accepted/attempts, "
+                  f"t={s.time:.4f}, dt={s.dt:.5f}, {el:.0f}s", flush=True)
+    wall = time.perf_counter() - t0
+    print(f"{accepted} accepted steps / {attempts} attempts in {wall:.1f}s "
+          f"= {accepted/max(wall,1e-9):.3f} steps/s wall; "
+          f"sim time reached t={s.time:.4f}", flush=True)
+    print(f"iters/accepted-step: min={min(iters_hist)} "
+          f"median={int(np.median(iters_hist))} max={max(iters_hist)}; "
+          f"dt: min={min(dt_hist):.5f} max={max(dt_hist):.5f}", flush=True)
+
+    # velocity-field parity vs the CPU oracle on the final state
+    import oracle
+    rng = np.random.default_rng(2)
+    pts = 0.35 * fx["nodes"][rng.integers(0, len(fx["nodes"]), 64)]
+    r_fib = s.fiber_nodes()
+    w = np.concatenate([f.quadrature_weights() for f in fibers])
+    fw = np.concatenate([(f.force_scale * f.xs).T for f in fibers])
+    dens = s.solution[s.fiber_sol_size:].reshape(-1, 3)
+    T = lambda a: torch.from_numpy(np.ascontiguousarray(a)).to(dev)
+    u_gpu = velocity_at_targets(
+        T(pts), 1.0,
+        fiber=dict(r_src=T(r_fib), forces=T(fw), weights=T(w)),
+        shell=dict(node_pos=T(fx["nodes"]),
+                   node_normal=T(np.ascontiguousarray(own["normals"])),
+                   density=T(dens))).cpu().numpy()
+    torch.cuda.synchronize()
+    u_cpu = oracle.stokeslet(r_fib, fw * w[:, None], pts, 1.0)
+    f_dl = 2.0 * np.einsum("ni,nj->nij", own["normals"], dens).reshape(-1, 9)
+    u_cpu += oracle.stresslet(fx["nodes"], f_dl, pts, 1.0)
+    rel = np.linalg.norm(u_gpu - u_cpu) / np.linalg.norm(u_cpu)
+    print(f"final-state velocity field at 64 interior targets: "
+          f"GPU vs CPU oracle rel = {rel:.3e}", flush=True)
+
+
+if __name__ == "__main__":
+    main()
