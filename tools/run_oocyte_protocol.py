@@ -33,6 +33,10 @@ def main():
     ap.add_argument("--steps", type=int, default=30,
                     help="accepted timesteps to run")
     ap.add_argument("--clearance", type=float, default=0.08)
+    ap.add_argument("--cap-frac", type=float, default=1.0,
+                    help="exclude seed nodes with |x| beyond this fraction "
+                         "of the envelope half-length (tip regions focus "
+                         "inward fibers onto the axis)")
     ap.add_argument("--dt", type=float, default=0.01)
     ap.add_argument("--tol", type=float, default=1e-8)
     ap.add_argument("--maxiter", type=int, default=1500)
@@ -57,7 +61,16 @@ def main():
           f"{time.perf_counter()-t0:.1f}s", flush=True)
     shell = Shell(fx["nodes"], own["normals"], A, M_inv)
 
-    fibers = place_fibers(fx, own["normals"], args.fibers, 32,
+    if args.cap_frac < 1.0:
+        # place from a cap-filtered seed set; the shell stays complete
+        half = 0.5 * float(fx["envelope_length"]) * float(fx["scale_factor"])
+        keep = np.abs(fx["nodes"][:, 0]) < args.cap_frac * half
+        seed_fix = {k: fx[k] for k in fx.files}
+        seed_fix["nodes"] = fx["nodes"][keep]
+        seed_normals = own["normals"][keep]
+    else:
+        seed_fix, seed_normals = fx, own["normals"]
+    fibers = place_fibers(seed_fix, seed_normals, args.fibers, 32,
                           clearance=args.clearance)
     print(f"placed {len(fibers)} fibers (clearance {args.clearance})",
           flush=True)
@@ -84,6 +97,18 @@ def main():
             if err <= 0.09:
                 s.dt = min(args.dt, s.dt * 1.2)
         else:
+            # locate the rejection hotspot: worst-error fiber
+            worst, we = None, 0.0
+            for f in s.fibers:
+                m = f.mats
+                xs = (2.0 / f.length) * f.x @ m["D_1_0"]
+                e = float(np.abs(np.linalg.norm(xs, axis=0) - 1.0).max())
+                if e > we:
+                    we, worst = e, f
+            print(f"  REJECT dt={s.dt:.5f} conv={info['converged']} "
+                  f"err={err:.2e} worst fiber minus-end "
+                  f"{np.round(worst.x[:, 0], 2)} plus-end "
+                  f"{np.round(worst.x[:, -1], 2)}", flush=True)
             s.dt *= 0.5
             s.restore()
             if s.dt < 1e-5:
