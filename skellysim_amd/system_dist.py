@@ -189,10 +189,17 @@ class DistributedSystemFD(SystemFD):
             if len(r_fib_all) else np.zeros_like(r_local)
         # per-fiber self subtraction on OWN fibers (local targets lead)
         if self.fibers and self.fibers[0].stokeslet is None:
-            pts = np.stack([f.x.T for f in self.fibers])
-            G = self.backend.self_stokeslet_batch(pts, self.eta)
-            for f, g in zip(self.fibers, G):
-                f.stokeslet = g
+            if self._uniform:
+                pts = np.stack([f.x.T for f in self.fibers])
+                G = self.backend.self_stokeslet_batch(pts, self.eta)
+                for f, g in zip(self.fibers, G):
+                    f.stokeslet = g
+            else:
+                # mixed discretizations: per-fiber batches of one (same
+                # fallback as SystemFD._fiber_flow)
+                for f in self.fibers:
+                    f.stokeslet = self.backend.self_stokeslet_batch(
+                        f.x.T[None], self.eta)[0]
         for f, a, b in self._fiber_node_slices():
             v_all[a:b] -= (f.stokeslet @ wf_local[a:b].reshape(-1)).reshape(f.n_nodes, 3)
 

@@ -349,3 +349,92 @@ def test_distributed_multistep_matches_single_process():
     for xd, f in zip(dist_xs, sys_.fibers):
         rel = np.linalg.norm(xd - f.x) / max(np.linalg.norm(f.x), 1e-30)
         assert rel < 1e-6, rel
+
+
+def _make_mixed_problem():
+    """Mixed-discretization fibers (24/32 nodes) — exercises the
+    per-fiber LU fallback on the distributed host preconditioner path
+    (the uniform reshape used to crash on these)."""
+    from skellysim_amd.fiber_fd import FiberFD
+    here = os.path.dirname(os.path.abspath(__file__))
+    fx = np.load(os.path.join(here, "golden", "periphery_sphere_192.npz"))
+    rng = np.random.default_rng(78)
+    fibers = []
+    for k, n in enumerate((24, 32, 24, 32)):
+        d = rng.uniform(-1, 1, 3)
+        x0 = rng.uniform(-0.25, 0.25, 3)
+        d /= np.linalg.norm(d)
+        s = np.linspace(0, 0.5, n)
+        x = x0[None, :] + s[:, None] * d[None, :]
+        fibers.append(FiberFD(x, length=0.5, bending_rigidity=2.5e-3, eta=1.0,
+                              minus_clamped=(k % 2 == 0), force_scale=-0.05))
+    U = np.array([0.05, 0.02, -0.04])
+    bg = lambda r: np.tile(U, (len(r), 1))
+    return fx, fibers, bg
+
+
+def _dist_mixed_worker(rank, world, init_file, q):
+    import torch.distributed as dist
+    from skellysim_amd.system_dist import DistributedSystemFD, distribute_fibers
+    from skellysim_amd.system_fd import Shell
+    from oracle_backend import OracleBackend
+    from skellysim_amd.sharded import shard_range
+
+    dist.init_process_group("gloo", init_method=f"file://{init_file}",
+                            rank=rank, world_size=world)
+    try:
+        fx, fibers, bg = _make_mixed_problem()
+        N = len(fx["nodes"])
+        a, b = shard_range(N, world, rank)
+        shell = Shell(fx["nodes"], fx["normals"],
+                      fx["stresslet_plus_complementary"][3 * a: 3 * b],
+                      fx["M_inv"][3 * a: 3 * b])
+        my_fibers = distribute_fibers(fibers, world, rank)
+        sys_ = DistributedSystemFD(my_fibers, eta=1.0, dt=0.05, shell=shell,
+                                   shell_rows=(a, b),
+                                   backend=OracleBackend(),
+                                   background_flow=bg)
+        info = sys_.solve(tol=1e-11, maxiter=300, restart=100)
+        q.put((rank, sys_.solution, sys_.fiber_sol_size, info["converged"]))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_distributed_mixed_discretization_matches_single_process():
+    with tempfile.TemporaryDirectory() as td:
+        init_file = os.path.join(td, "pg")
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        procs = [ctx.Process(target=_dist_mixed_worker,
+                             args=(r, WORLD, init_file, q))
+                 for r in range(WORLD)]
+        for p in procs:
+            p.start()
+        results = {}
+        for _ in range(WORLD):
+            rank, sol, fib_size, conv = q.get(timeout=250)
+            assert conv
+            results[rank] = (sol, fib_size)
+        for p in procs:
+            p.join(timeout=60)
+            assert p.exitcode == 0
+
+    from skellysim_amd.system_fd import SystemFD, Shell
+    from oracle_backend import OracleBackend
+    fx, fibers, bg = _make_mixed_problem()
+    shell = Shell(fx["nodes"], fx["normals"],
+                  fx["stresslet_plus_complementary"], fx["M_inv"])
+    sys_ = SystemFD(fibers, eta=1.0, dt=0.05, shell=shell,
+                    backend=OracleBackend(), background_flow=bg)
+    assert sys_.solve(tol=1e-11, maxiter=300, restart=100)["converged"]
+
+    fib_parts, shell_parts = [], []
+    for r in range(WORLD):
+        sol, fib_size = results[r]
+        fib_parts.append(sol[:fib_size])
+        shell_parts.append(sol[fib_size:])
+    assembled = np.concatenate(fib_parts + shell_parts)
+    rel = np.linalg.norm(assembled - sys_.solution) / \
+        np.linalg.norm(sys_.solution)
+    assert rel < 1e-9, rel
