@@ -43,6 +43,14 @@ extern "C" {
 
 /* ---- library / device management ---- */
 
+/* Force the persistent (grow-only hipMalloc) split-K workspace on/off at
+ * runtime, overriding the SKELLY_PERSISTENT_WS env var (on=1/0; -1 not
+ * accepted — call once). Required before hipGraph capture of launches
+ * that take the split path: the async-mempool workspace cannot be
+ * recorded, while the persistent one allocates only during uncaptured
+ * warmup. */
+void skelly_set_persistent_ws(int on);
+
 /* Human-readable build id ("skelly-hip <ver> gfx950"). */
 const char *skelly_hip_version(void);
 

@@ -51,6 +51,8 @@ def _bind(lib):
     lib.skelly_oseen_tensor_batched_device.argtypes = [pv, pv, _LL, _LL, ctypes.c_double,
                                                         ctypes.c_double, ctypes.c_double, pv]
     lib.skelly_fp64_peak_tflops.argtypes = [ctypes.POINTER(ctypes.c_double)]
+    lib.skelly_set_persistent_ws.argtypes = [ctypes.c_int]
+    lib.skelly_set_persistent_ws.restype = None
 
 
 def lib():

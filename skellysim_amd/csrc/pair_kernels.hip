@@ -372,7 +372,11 @@ namespace skelly {
  * alloc/free pairs are the prime suspect for the sync-cadence corruption
  * documented in gmres.py / DESIGN.md — flip this on to test that
  * hypothesis. */
+static int g_persistent_ws_override = -1; /* -1 = follow env var */
+
 static bool persistent_ws_enabled() {
+    if (g_persistent_ws_override >= 0)
+        return g_persistent_ws_override != 0;
     static const bool on = [] {
         const char *e = getenv("SKELLY_PERSISTENT_WS");
         return e && atoi(e) != 0;
@@ -693,3 +697,12 @@ hipError_t run_fp64_peak(double *out_tflops) {
 }
 
 } // namespace skelly
+
+/* Runtime override of the persistent split-K workspace (see
+ * persistent_ws_enabled above): hipGraph capture of the GMRES iteration
+ * (gmres.py use_graph) requires it — hipMallocAsync nodes cannot be
+ * recorded reliably, and the grow-only hipMalloc happens during the
+ * UNCAPTURED warmup pass, so capture itself allocates nothing. */
+extern "C" void skelly_set_persistent_ws(int on) {
+    skelly::g_persistent_ws_override = on;
+}

@@ -38,9 +38,49 @@ def _make_reduce(distributed, group=None):
     return allreduce
 
 
+def _arnoldi_body(matvec, precond, V, Hstage, idx_k, idx_k1):
+    """One shape-invariant Arnoldi iteration against the FULL basis: rows
+    of V beyond the current k are zero, so their ICGS contributions vanish
+    exactly; idx_k/idx_k1 are 1-element device index tensors whose VALUES
+    are read at replay time."""
+    vk = V.index_select(0, idx_k).reshape(-1)
+    w = matvec(precond(vk))
+    h1 = V @ w
+    w = w - V.T @ h1
+    h2 = V @ w
+    w = w - V.T @ h2
+    hk1 = torch.sqrt(torch.dot(w, w))
+    V.index_copy_(0, idx_k1, (w / hk1).unsqueeze(0))
+    Hstage.index_copy_(0, idx_k,
+                       torch.cat([h1 + h2, hk1.reshape(1)]).unsqueeze(0))
+
+
+def _capture_arnoldi(matvec, precond, V, Hstage, idx_k, idx_k1):
+    """Warm up (materializes rocBLAS/extension workspaces — including the
+    persistent split-K buffers, which must pre-exist because capture may
+    not allocate), then record one iteration as a hipGraph. The warmup
+    corrupts V[1]/Hstage[0]; the CALLER must re-zero V[1:]/Hstage after."""
+    from . import _native
+    try:
+        _native.lib().skelly_set_persistent_ws(1)
+    except (RuntimeError, AttributeError):
+        pass  # CPU-only environment; capture will fail loudly if reached
+    g = torch.cuda.CUDAGraph()
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        for _ in range(2):
+            _arnoldi_body(matvec, precond, V, Hstage, idx_k, idx_k1)
+    torch.cuda.current_stream().wait_stream(s)
+    torch.cuda.synchronize()
+    with torch.cuda.graph(g):
+        _arnoldi_body(matvec, precond, V, Hstage, idx_k, idx_k1)
+    return g
+
+
 def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
           x0=None, distributed=False, group=None, callback=None,
-          sync_cadence=None):
+          sync_cadence=None, use_graph=None):
     """Solve A x = b with right-preconditioned GMRES(restart).
 
     matvec(v) -> A v ; precond(v) -> M^-1 v (right preconditioner: solves
@@ -61,10 +101,27 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
     preconditioner — with the trsm-based LU solves (batched.py, now the
     default) cadence 8 is bitwise equal to cadence 1 on device at config-5
     scale. Override via the SKELLY_GMRES_SYNC_CADENCE env var.
+
+    use_graph: capture ONE Arnoldi iteration (matvec∘precond + two-pass
+    ICGS + normalize + Hessenberg staging) as a hipGraph and replay it per
+    iteration — removes the per-iteration kernel-launch/Python-dispatch
+    overhead (~1.4 ms ICGS+glue + ~1.6 ms matvec glue measured at config-4
+    shape, tools/prof_iter.py). Shape-invariance comes from
+    orthogonalizing against the FULL (m+1)-row basis every iteration
+    (rows beyond the current k are zero, so the extra terms vanish
+    exactly) with device-index tensors selecting the read/write rows —
+    the cost is (m+1)·n GEMV traffic per iteration instead of (k+1)·n.
+    Default (None) = the SKELLY_HIPGRAPH env var (off); requires CUDA
+    tensors, non-distributed. Results are ULP-equivalent, not bitwise
+    (GEMV reduction order differs with the padded rows).
     """
     if sync_cadence is None:
         import os
         sync_cadence = int(os.environ.get("SKELLY_GMRES_SYNC_CADENCE", "8"))
+    if use_graph is None:
+        import os
+        use_graph = os.environ.get("SKELLY_HIPGRAPH", "0") == "1"
+    use_graph = bool(use_graph) and b.is_cuda and not distributed
     if precond is None:
         precond = lambda v: v
     reduce_ = _make_reduce(distributed, group)
@@ -84,6 +141,15 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
     converged = False
     true_resid = None
 
+    gstate = None
+    if use_graph:
+        m_fix = min(restart, maxiter)
+        V_g = torch.zeros((m_fix + 1, n), dtype=b.dtype, device=b.device)
+        Hstage = torch.zeros((m_fix, m_fix + 2), dtype=b.dtype,
+                             device=b.device)
+        idx_k = torch.zeros(1, dtype=torch.long, device=b.device)
+        idx_k1 = torch.ones(1, dtype=torch.long, device=b.device)
+
     while total_iters < maxiter and not converged:
         r = b - matvec(x)
         beta = norm(r)
@@ -99,8 +165,20 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
         # coalesced paths — the (n, m+1) layout made V_k h a strided
         # column-slice gemv that dominated whole solves (53% of config-5
         # GPU time at 278 GB/s; profiles/components_r01.md)
-        V = torch.zeros((m + 1, n), dtype=b.dtype, device=b.device)
-        V[0] = r / beta
+        if use_graph:
+            V = V_g
+            V.zero_()
+            V[0] = r / beta
+            if gstate is None:
+                idx_k.fill_(0)
+                idx_k1.fill_(1)
+                gstate = _capture_arnoldi(matvec, precond, V, Hstage,
+                                          idx_k, idx_k1)
+                V[1:].zero_()  # warmup corrupted V[1]
+            Hstage.zero_()
+        else:
+            V = torch.zeros((m + 1, n), dtype=b.dtype, device=b.device)
+            V[0] = r / beta
         # small dense Hessenberg/Givens state in numpy: scalar torch-CPU ops
         # cost a fork-join on many-core hosts
         H = np.zeros((m + 1, m))
@@ -113,30 +191,45 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
         pending = []  # deferred (device-side) Hessenberg columns
         stop = False
         for k in range(m):
-            w = matvec(precond(V[k]))
-            # ICGS: two classical Gram-Schmidt passes (Belos "ICGS",
-            # solver_hydro.cpp:72), each as one fused dot-block + update.
-            # Scalars stay on device; host transfers (full-stream syncs)
-            # are batched every sync_cadence iterations.
-            Vk = V[: k + 1]
-            hcol_dev = None
-            for _ in range(2):
-                h = reduce_(Vk @ w)
-                w = w - Vk.T @ h
-                hcol_dev = h if hcol_dev is None else hcol_dev + h
-            hk1_dev = torch.sqrt(reduce_(torch.dot(w, w)))
-            # device-side normalize without a host read; if hk1 == 0 the
-            # row is never consumed (processing below discards past it)
-            V[k + 1] = w / hk1_dev
-            pending.append(torch.cat([hcol_dev.reshape(-1),
-                                      hk1_dev.reshape(1)]))
+            if use_graph:
+                idx_k.fill_(k)
+                idx_k1.fill_(k + 1)
+                gstate.replay()
+                pending.append(k)  # marker; columns live in Hstage
+            else:
+                w = matvec(precond(V[k]))
+                # ICGS: two classical Gram-Schmidt passes (Belos "ICGS",
+                # solver_hydro.cpp:72), each as one fused dot-block +
+                # update. Scalars stay on device; host transfers
+                # (full-stream syncs) are batched every sync_cadence
+                # iterations.
+                Vk = V[: k + 1]
+                hcol_dev = None
+                for _ in range(2):
+                    h = reduce_(Vk @ w)
+                    w = w - Vk.T @ h
+                    hcol_dev = h if hcol_dev is None else hcol_dev + h
+                hk1_dev = torch.sqrt(reduce_(torch.dot(w, w)))
+                # device-side normalize without a host read; if hk1 == 0
+                # the row is never consumed (processing below discards
+                # past it)
+                V[k + 1] = w / hk1_dev
+                pending.append(torch.cat([hcol_dev.reshape(-1),
+                                          hk1_dev.reshape(1)]))
             if len(pending) < max(1, sync_cadence) and k != m - 1:
                 continue
 
             # flush: ONE host transfer, then sequential Givens bookkeeping
             # for the batched columns
-            scalars = torch.cat(pending).cpu().numpy()
             k0 = k + 1 - len(pending)
+            if use_graph:
+                rows = Hstage[k0: k + 1].cpu().numpy()
+                scalars = np.concatenate(
+                    [np.concatenate([rows[i][: k0 + i + 2 - 1],
+                                     rows[i][m_fix + 1: m_fix + 2]])
+                     for i in range(len(rows))])
+            else:
+                scalars = torch.cat(pending).cpu().numpy()
             off = 0
             for kk in range(k0, k + 1):
                 col_scalars = scalars[off: off + kk + 2]
