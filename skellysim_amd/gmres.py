@@ -73,7 +73,11 @@ def _capture_arnoldi(matvec, precond, V, Hstage, idx_k, idx_k1):
             _arnoldi_body(matvec, precond, V, Hstage, idx_k, idx_k1)
     torch.cuda.current_stream().wait_stream(s)
     torch.cuda.synchronize()
-    with torch.cuda.graph(g):
+    # capture on the SAME stream the warmup ran on: the extension's
+    # persistent split-K workspace is cached per stream, so a different
+    # capture stream would miss the cache and hipMalloc inside capture
+    # (-> hipErrorStreamCaptureInvalidated)
+    with torch.cuda.graph(g, stream=s):
         _arnoldi_body(matvec, precond, V, Hstage, idx_k, idx_k1)
     return g
 
