@@ -107,17 +107,20 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
     scale. Override via the SKELLY_GMRES_SYNC_CADENCE env var.
 
     use_graph: capture ONE Arnoldi iteration (matvec∘precond + two-pass
-    ICGS + normalize + Hessenberg staging) as a hipGraph and replay it per
-    iteration — removes the per-iteration kernel-launch/Python-dispatch
-    overhead (~1.4 ms ICGS+glue + ~1.6 ms matvec glue measured at config-4
-    shape, tools/prof_iter.py). Shape-invariance comes from
-    orthogonalizing against the FULL (m+1)-row basis every iteration
-    (rows beyond the current k are zero, so the extra terms vanish
-    exactly) with device-index tensors selecting the read/write rows —
-    the cost is (m+1)·n GEMV traffic per iteration instead of (k+1)·n.
-    Default (None) = the SKELLY_HIPGRAPH env var (off); requires CUDA
-    tensors, non-distributed. Results are ULP-equivalent, not bitwise
-    (GEMV reduction order differs with the padded rows).
+    ICGS + normalize + Hessenberg staging) as a hipGraph and replay it
+    per iteration. Shape-invariance comes from orthogonalizing against
+    the FULL (m+1)-row basis every iteration (rows beyond the current k
+    are zero, so the extra terms vanish exactly) with device-index
+    tensors selecting the read/write rows — the cost is (m+1)·n GEMV
+    traffic per iteration instead of (k+1)·n. MEASURED (round 2,
+    tools/prof_iter.py at config-4 shape): 7.31 vs 6.78 ms/iteration —
+    the sync-cadence-8 batching already keeps the host far enough ahead
+    that launch overhead is hidden, so the graph only adds the padded
+    GEMV traffic; it stays OPT-IN (SKELLY_HIPGRAPH=1; requires CUDA
+    tensors, non-distributed), validated for correctness on GPU
+    (tests/test_gpu_graph.py: same solution and iteration count as the
+    eager path). Results are ULP-equivalent, not bitwise (GEMV reduction
+    order differs with the padded rows).
     """
     if sync_cadence is None:
         import os
