@@ -101,10 +101,12 @@ def gmres(matvec, b, precond=None, tol=1e-10, maxiter=200, restart=30,
     (tests/test_gmres.py). DEFAULT 8 (~20% faster per iteration at
     config-5 scale): safe since the round-2 experiment matrix
     (profiles/cadence_matrix_r02.md) isolated the deep-queue corruption
-    that cadence>1 used to trigger to magma's torch.linalg.lu_solve in the
-    preconditioner — with the trsm-based LU solves (batched.py, now the
-    default) cadence 8 is bitwise equal to cadence 1 on device at config-5
-    scale. Override via the SKELLY_GMRES_SYNC_CADENCE env var.
+    that cadence>1 used to trigger to magma's torch.linalg.lu_solve in
+    the preconditioner — with magma out of the iteration (batched.py
+    SKELLY_LU_MODE=inv default: explicit probed inverse, one bmm per
+    apply) deep cadences are stable; the trsm mode was measured bitwise
+    equal to cadence 1 at config-5 scale. Override via the
+    SKELLY_GMRES_SYNC_CADENCE env var.
 
     use_graph: capture ONE Arnoldi iteration (matvec∘precond + two-pass
     ICGS + normalize + Hessenberg staging) as a hipGraph and replay it
