@@ -127,6 +127,9 @@ class SystemFD:
         self.periphery_shape = periphery_shape
         self.periphery_binding = periphery_binding
         self.dynamic_instability = dynamic_instability
+        # fiber-fiber steric repulsion (ENGINE EXTENSION, default off):
+        # dict(f_0=, l_0=, d_max=) — see _fiber_fiber_repulsion
+        self.steric_interaction = None
         self.rng = np.random.default_rng(seed)  # params.seed default 130319
         # point/background sources (sources.py); simulation clock for their
         # lifetimes (properties.time — advanced by run())
@@ -329,12 +332,17 @@ class SystemFD:
             pi = self.periphery_interaction
             for f, a, b in self._fiber_node_slices():
                 ext[a:b] = f.periphery_repulsion(**pi).T
+        # fiber-fiber steric repulsion (ENGINE EXTENSION, opt-in; see
+        # _fiber_fiber_repulsion)
+        if self.steric_interaction is not None and nf_nodes:
+            ext += self._fiber_fiber_repulsion(**self.steric_interaction)
 
         # v_all: flow induced by the external (periphery) forces
         # (fc_->flow(r_all, external_force_fibers), system.cpp:425)
         # + background flow; motor forces enter only the RHS
         v_all = np.zeros_like(r_all)
-        if self.periphery_interaction is not None and nf_nodes:
+        if nf_nodes and (self.periphery_interaction is not None
+                         or self.steric_interaction is not None):
             v_all += self._fiber_flow(r_all, ext)
         if self.background_flow is not None:
             v_all += self.background_flow(r_all)
@@ -808,6 +816,51 @@ class SystemFD:
             b.step(self.dt, self.solution[a:bb])
         self.repin_to_bodies()
         return info
+
+    def _fiber_fiber_repulsion(self, f_0=20.0, l_0=0.05, d_max=None,
+                               fiber_radius=0.0125):
+        """Pairwise fiber-fiber steric repulsion — an ENGINE EXTENSION,
+        default OFF (set system.steric_interaction = dict(f_0=, l_0=,
+        d_max=)). The reference implements steric interaction only
+        against the periphery (periphery.cpp:140-162); at its own dense
+        oocyte packing (examples/oocyte, ~0.1 fiber spacing) near-contact
+        fiber pairs develop quadrature-singular velocities that reject
+        timesteps at any dt (profiles/oocyte_r02.md). This applies the
+        SAME exponential force law pairwise between nodes of different
+        fibers: f = f_0 * exp(-gap/l_0) * unit(dx), equal and opposite,
+        gap = center distance - 2*fiber_radius (clamped >= 0), cutoff
+        d_max (default 5*l_0 + 2*fiber_radius)."""
+        from scipy.spatial import cKDTree
+        import torch.distributed as dist
+        if dist.is_available() and dist.is_initialized() \
+                and dist.get_world_size() > 1:
+            raise NotImplementedError(
+                "fiber-fiber steric repulsion needs a global neighbor "
+                "search; not wired into the distributed prep yet")
+        pts = self.fiber_nodes()
+        out = np.zeros_like(pts)
+        if d_max is None:
+            d_max = 5.0 * l_0 + 2.0 * fiber_radius
+        fid = np.concatenate([np.full(f.n_nodes, i)
+                              for i, f in enumerate(self.fibers)])
+        pairs = cKDTree(pts).query_pairs(r=d_max, output_type="ndarray")
+        if len(pairs) == 0:
+            return out
+        i, j = pairs[:, 0], pairs[:, 1]
+        keep = fid[i] != fid[j]
+        i, j = i[keep], j[keep]
+        if len(i) == 0:
+            return out
+        dx = pts[i] - pts[j]
+        d = np.linalg.norm(dx, axis=1)
+        pos = d > 0
+        i, j, dx, d = i[pos], j[pos], dx[pos], d[pos]
+        gap = np.maximum(d - 2.0 * fiber_radius, 0.0)
+        mag = f_0 * np.exp(-gap / l_0) / d
+        fvec = mag[:, None] * dx
+        np.add.at(out, i, fvec)
+        np.add.at(out, j, -fvec)
+        return out
 
     def repin_to_bodies(self):
         """Translate every body-attached fiber so its minus end coincides
