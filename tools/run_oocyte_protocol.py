@@ -40,6 +40,8 @@ def main():
     ap.add_argument("--dt", type=float, default=0.01)
     ap.add_argument("--tol", type=float, default=1e-8)
     ap.add_argument("--maxiter", type=int, default=1500)
+    ap.add_argument("--steric-f0", type=float, default=20.0)
+    ap.add_argument("--steric-l0", type=float, default=0.05)
     ap.add_argument("--steric", action="store_true",
                     help="enable the engine's fiber-fiber steric repulsion "
                          "(reference force law f_0=20, l_0=0.05 applied "
@@ -82,9 +84,9 @@ def main():
     s = SystemFD(fibers, eta=1.0, dt=args.dt, shell=shell,
                  backend=HipBackend())
     if args.steric:
-        s.steric_interaction = dict(f_0=20.0, l_0=0.05)
-        print("fiber-fiber steric repulsion ON (f_0=20, l_0=0.05)",
-              flush=True)
+        s.steric_interaction = dict(f_0=args.steric_f0, l_0=args.steric_l0)
+        print(f"fiber-fiber steric repulsion ON (f_0={args.steric_f0}, "
+              f"l_0={args.steric_l0})", flush=True)
     print(f"solution size: {s.fiber_sol_size + s.shell_sol_size}", flush=True)
 
     accepted = 0
