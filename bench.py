@@ -129,7 +129,11 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     dist = None
-    if world > 1:
+    # initialize the process group whenever a launcher provided WORLD_SIZE
+    # (even world 1): a torchrun world-1 run then exercises the full RCCL
+    # init/barrier/allreduce path on one GPU — the same code the driver's
+    # 2/4/8-GPU SCALE launch takes
+    if "WORLD_SIZE" in os.environ:
         import torch.distributed as dist_mod
         dist = dist_mod
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
