@@ -59,11 +59,11 @@ def _check(sigma, pin1, pin2, growing):
     assert rel < 1e-6, (p1, p2, rel)     # the reference's own gate
 
 
-@pytest.mark.timeout(600)
+@pytest.mark.timeout(1200)
 def test_clamped_buckling_sigma72_subcritical():
     _check(72.0, 0.08844356, 0.05563314, growing=False)
 
 
-@pytest.mark.timeout(600)
+@pytest.mark.timeout(1200)
 def test_clamped_buckling_sigma80_supercritical():
     _check(80.0, 0.09575812, 0.13564472, growing=True)

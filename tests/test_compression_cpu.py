@@ -21,7 +21,7 @@ from compression_common import (build_system, run_protocol,
 from oracle_backend import OracleBackend
 
 
-@pytest.mark.timeout(900)
+@pytest.mark.timeout(2400)
 def test_compression_regression_oracle():
     be = OracleBackend()
     s = build_system(be, device=False)
